@@ -1,0 +1,90 @@
+"""The 216-cell configuration grid, declaratively.
+
+The reference builds its grid out of live sklearn/imblearn estimator objects
+(reference experiment.py:73-100).  Here the grid is pure data: each axis maps a
+display key to a spec understood by the engine.  The AXIS ORDER and KEY ORDER
+are a compatibility contract — `scores.pkl` is keyed by the tuple of display
+keys in this order, and downstream consumers (top-10 tables, comparison
+tables, shap configs) index it that way.
+
+Axes (2 x 2 x 3 x 6 x 3 = 216 cells):
+  0. flaky type   : NOD -> label FLAKY(2), OD -> label OD_FLAKY(1)
+  1. feature set  : Flake16 (all 16 columns) | FlakeFlagger (7 columns)
+  2. preprocessing: None | Scaling (StandardScaler) | PCA (Scaler -> full PCA)
+  3. balancing    : None | Tomek Links | SMOTE | ENN | SMOTE ENN | SMOTE Tomek
+  4. model        : Extra Trees | Random Forest | Decision Tree  (100/100/1
+                    estimators, sklearn 1.0.2 defaults, random_state=0)
+"""
+
+import itertools
+
+from .constants import FEATURE_NAMES, FLAKEFLAGGER_COLUMNS, FLAKY, OD_FLAKY
+
+FLAKY_TYPE_AXIS = {
+    "NOD": FLAKY,
+    "OD": OD_FLAKY,
+}
+
+FEATURE_SET_AXIS = {
+    "Flake16": tuple(range(len(FEATURE_NAMES))),
+    "FlakeFlagger": FLAKEFLAGGER_COLUMNS,
+}
+
+PREPROCESSING_AXIS = {
+    "None": None,
+    "Scaling": "scale",
+    "PCA": "scale+pca",       # StandardScaler -> PCA(all components), like the
+                              # reference's Pipeline (experiment.py:85)
+}
+
+BALANCING_AXIS = {
+    "None": None,
+    "Tomek Links": "tomek",
+    "SMOTE": "smote",
+    "ENN": "enn",
+    "SMOTE ENN": "smote+enn",
+    "SMOTE Tomek": "smote+tomek",
+}
+
+MODEL_AXIS = {
+    "Extra Trees": {"kind": "extra_trees", "n_estimators": 100, "bootstrap": False},
+    "Random Forest": {"kind": "random_forest", "n_estimators": 100, "bootstrap": True},
+    "Decision Tree": {"kind": "decision_tree", "n_estimators": 1, "bootstrap": False},
+}
+
+CONFIG_GRID = (
+    FLAKY_TYPE_AXIS,
+    FEATURE_SET_AXIS,
+    PREPROCESSING_AXIS,
+    BALANCING_AXIS,
+    MODEL_AXIS,
+)
+
+# The two fixed shap-stage configs (reference experiment.py:524-525).
+SHAP_CONFIGS = (
+    ("NOD", "Flake16", "Scaling", "SMOTE Tomek", "Extra Trees"),
+    ("OD", "Flake16", "Scaling", "SMOTE", "Random Forest"),
+)
+
+
+def iter_config_keys():
+    """All 216 config-key tuples in itertools.product order — the same
+    enumeration order as the reference's write_scores (experiment.py:494)."""
+    return itertools.product(*[d.keys() for d in CONFIG_GRID])
+
+
+def resolve(config_keys):
+    """Config-key tuple -> (flaky_label, feature_set, preproc, balancing, model)."""
+    return [CONFIG_GRID[i][k] for i, k in enumerate(config_keys)]
+
+
+def cell_cost_estimate(config_keys):
+    """Rough relative cost of a cell, for load balancing across ranks.
+
+    Forest cells (100 trees) dominate; SMOTE-family balancers roughly double
+    the training-set size (minority oversampled to parity).
+    """
+    _, _, _, balancing, model = config_keys
+    n_trees = MODEL_AXIS[model]["n_estimators"]
+    balance_mult = 2.0 if "SMOTE" in balancing else 1.0
+    return n_trees * balance_mult + 1.0

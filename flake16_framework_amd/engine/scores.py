@@ -1,0 +1,158 @@
+"""The `scores` stage: evaluate the 216-cell grid, write scores.pkl.
+
+Output contract (reference experiment.py:446-501):
+  scores.pkl = { (k0..k4): [t_train, t_test, scores, scores_total] }
+    scores       = {proj: [FP, FN, TP, P, R, F]}  (all projects in dataset)
+    scores_total = [FP, FN, TP, P, R, F]
+    t_train/t_test = MEAN seconds per fold over the 10 folds
+  quirks preserved: preprocessing is fit on the FULL dataset before the CV
+  split; true negatives are skipped; P/R/F are None on zero denominators.
+
+Execution model (MI355X-first, nothing like the reference's process pool):
+  cells are sharded across ranks (one rank per GPU); each rank evaluates its
+  cells device-resident — fold-batched balancing, binning, forest fit,
+  prediction and confusion — and the per-cell result blobs are combined with
+  one RCCL all-reduce at the end (parallel/comm.py).  The 'ref' backend runs
+  the identical algorithm in numpy on CPU.
+"""
+
+import pickle
+import time
+
+import numpy as np
+
+from ..balance import apply_balancing
+from ..configgrid import iter_config_keys, resolve
+from ..constants import SCORES_FILE
+from ..dataset.tests_io import load_feat_lab_proj
+from ..models.binning import bin_codes, compute_bin_cuts
+from ..models.forest_ref import fit_forest, params_for_model, predict_forest
+from ..preprocess import apply_preprocessing
+from .folds import stratified_kfold_split
+from .metrics import finalize_scores
+
+GLOBAL_SEED = 0
+N_FOLDS = 10
+
+
+def job_ids_for(cell_idx, fold):
+    """Philox key-ids for one (cell, fold): (balance_k1, tree_job_base).
+    Tree t of this fold uses k1 = tree_job_base + t (t < 128)."""
+    ctx = cell_idx * N_FOLDS + fold
+    return ctx, ctx * 128
+
+
+def evaluate_cell_ref(config_keys, cell_idx, tests=None, tests_file=None,
+                      seed=GLOBAL_SEED):
+    """Evaluate one grid cell with the numpy reference backend.
+
+    Returns [t_train, t_test, scores, scores_total] (the scores.pkl value).
+    """
+    flaky_label, feature_set, preproc, balancing, model = resolve(config_keys)
+    kwargs = {"tests": tests} if tests is not None else {"tests_file": tests_file}
+    features, labels, projects = load_feat_lab_proj(flaky_label, feature_set,
+                                                    **kwargs)
+
+    X = apply_preprocessing(features, preproc).astype(np.float32)
+    labels = labels.astype(np.uint8)
+
+    # Full-dataset bin cuts (mirrors the full-data preprocessing-fit quirk).
+    cuts = compute_bin_cuts(X)
+    codes_all = bin_codes(X, cuts)
+    params = params_for_model(model, seed=seed)
+
+    t_train = t_test = 0.0
+    scores = {proj: [0] * 6 for proj in projects}
+    scores_total = [0] * 6
+
+    for i, (train, test) in enumerate(
+            stratified_kfold_split(labels, n_splits=N_FOLDS,
+                                   random_state=seed)):
+        bal_k1, job_base = job_ids_for(cell_idx, i)
+        Xb, yb = apply_balancing(X[train], labels[train], balancing,
+                                 seed, bal_k1)
+        codes_tr = bin_codes(Xb, cuts)
+
+        t0 = time.time()
+        forest = fit_forest(codes_tr, yb, params, job_base=job_base, cuts=cuts)
+        t_train += time.time() - t0
+
+        t0 = time.time()
+        preds = predict_forest(forest, codes_all[test])
+        t_test += time.time() - t0
+
+        y_test = labels[test]
+        projects_test = projects[test]
+        for j in range(len(test)):
+            k = int(2 * y_test[j] + preds[j]) - 1
+            if k == -1:
+                continue
+            scores[projects_test[j]][k] += 1
+            scores_total[k] += 1
+
+    finalize_scores(scores, scores_total)
+    return [t_train / N_FOLDS, t_test / N_FOLDS, scores, scores_total]
+
+
+def run_scores(tests_file=None, tests=None, backend="auto", cells=None,
+               progress=None, seed=GLOBAL_SEED):
+    """Evaluate `cells` (an iterable of (cell_idx, config_keys); default all
+    216) and return {config_keys: [t_train, t_test, scores, scores_total]}.
+    """
+    all_cells = list(enumerate(iter_config_keys()))
+    if cells is not None:
+        wanted = set(cells)
+        all_cells = [(i, k) for i, k in all_cells if i in wanted]
+
+    if backend == "auto":
+        backend = _auto_backend()
+
+    out = {}
+    t_start = time.time()
+    for n_done, (cell_idx, config_keys) in enumerate(all_cells):
+        if backend == "hip":
+            from .hip_cell import evaluate_cell_hip
+            out[config_keys] = evaluate_cell_hip(
+                config_keys, cell_idx, tests=tests, tests_file=tests_file,
+                seed=seed)
+        else:
+            out[config_keys] = evaluate_cell_ref(
+                config_keys, cell_idx, tests=tests, tests_file=tests_file,
+                seed=seed)
+        if progress:
+            progress(n_done + 1, len(all_cells), time.time() - t_start,
+                     ", ".join(config_keys))
+    return out
+
+
+def _auto_backend():
+    try:
+        import torch
+        if torch.cuda.is_available():
+            return "hip"
+    except ImportError:
+        pass
+    return "ref"
+
+
+def write_scores(tests_file=None, scores_file=SCORES_FILE, backend="auto"):
+    """Full 216-cell sweep (sharded across ranks if distributed is
+    initialized) -> scores.pkl on rank 0."""
+    from ..parallel import comm
+
+    rank, world = comm.rank_world()
+    my_cells = comm.shard_cells(world, rank)
+
+    def progress(done, total, elapsed, name):
+        eta = elapsed / done * (total - done)
+        print(f"[rank {rank}] {done}/{total} {name} "
+              f"({elapsed:.0f}s elapsed, eta {eta:.0f}s)", flush=True)
+
+    result = run_scores(tests_file=tests_file, backend=backend,
+                        cells=my_cells, progress=progress)
+    result = comm.gather_scores(result)
+
+    if rank == 0:
+        with open(scores_file, "wb") as fd:
+            pickle.dump(result, fd)
+    return result

@@ -1,0 +1,135 @@
+"""Distributed execution: one rank per GPU, RCCL (= the "nccl" backend of
+torch.distributed on ROCm) over xGMI; gloo on CPU for tests.
+
+The reference's only transport is a multiprocessing.Pool result queue
+(experiment.py:191-211, 496).  Here the 216 grid cells are sharded across
+ranks cost-balanced, each rank evaluates its cells device-resident, and the
+per-cell result blobs are combined with ONE all-reduce at the end: payloads
+are tiny (per-cell confusion counts, ~70 KB for the whole grid), so latency
+dominates and a single fused collective beats any chatter.
+"""
+
+import os
+
+import numpy as np
+
+from ..configgrid import cell_cost_estimate, iter_config_keys
+from ..engine.metrics import finalize_scores
+
+_N_CONFUSION = 3  # FP, FN, TP
+
+
+def dist():
+    import torch.distributed as d
+    return d
+
+
+def is_initialized():
+    try:
+        return dist().is_initialized()
+    except Exception:
+        return False
+
+
+def rank_world():
+    if is_initialized():
+        return dist().get_rank(), dist().get_world_size()
+    return 0, 1
+
+
+def init_from_env(backend=None):
+    """Initialize the process group from torchrun env vars if present."""
+    if is_initialized() or "RANK" not in os.environ:
+        return rank_world()
+    import torch
+    if backend is None:
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+    if backend == "nccl":
+        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
+    dist().init_process_group(backend=backend)
+    return rank_world()
+
+
+def shard_cells(world, rank, n_cells=None):
+    """Cost-balanced static shard of the grid cells: greedy assignment of
+    cells (heaviest first) to the currently lightest rank.  Deterministic.
+    Returns the sorted list of cell indices owned by `rank`."""
+    keys = list(iter_config_keys())
+    if n_cells is not None:
+        keys = keys[:n_cells]
+    costs = [cell_cost_estimate(k) for k in keys]
+    order = sorted(range(len(keys)), key=lambda i: (-costs[i], i))
+
+    load = [0.0] * world
+    owner = [0] * len(keys)
+    for i in order:
+        r = min(range(world), key=lambda j: (load[j], j))
+        owner[i] = r
+        load[r] += costs[i]
+
+    return sorted(i for i in range(len(keys)) if owner[i] == rank)
+
+
+def _pack(result, cell_order, projects):
+    """result {config_keys: [t_train, t_test, scores, scores_total]} ->
+    float64 tensor [n_cells, 2 + (n_proj+1)*3]."""
+    n_proj = len(projects)
+    buf = np.zeros((len(cell_order), 2 + (n_proj + 1) * _N_CONFUSION))
+    for ci, keys in enumerate(cell_order):
+        if keys not in result:
+            continue
+        t_train, t_test, scores, scores_total = result[keys]
+        buf[ci, 0] = t_train
+        buf[ci, 1] = t_test
+        for pi, proj in enumerate(projects):
+            buf[ci, 2 + pi * 3: 2 + pi * 3 + 3] = scores[proj][:3]
+        buf[ci, 2 + n_proj * 3:] = scores_total[:3]
+    return buf
+
+
+def _unpack(buf, cell_order, projects):
+    n_proj = len(projects)
+    out = {}
+    for ci, keys in enumerate(cell_order):
+        scores = {}
+        for pi, proj in enumerate(projects):
+            scores[proj] = [int(v) for v in buf[ci, 2 + pi * 3: 2 + pi * 3 + 3]]
+        scores_total = [int(v) for v in buf[ci, 2 + n_proj * 3:]]
+        finalize_scores(scores, scores_total)
+        out[keys] = [float(buf[ci, 0]), float(buf[ci, 1]), scores, scores_total]
+    return out
+
+
+def gather_scores(result):
+    """Combine per-rank partial results into the full scores dict on every
+    rank via one all-reduce(SUM) over a flat tensor.  world==1: passthrough.
+    """
+    rank, world = rank_world()
+    if world == 1:
+        return result
+
+    import torch
+    d = dist()
+
+    cell_order = list(iter_config_keys())
+    # Every evaluated cell carries the full project list in dataset order.
+    projects = None
+    for v in result.values():
+        projects = list(v[2].keys())
+        break
+    # All ranks must agree on the project list; broadcast from rank 0's copy
+    # is unnecessary since all load the same tests.json, but a rank could in
+    # principle own zero cells — exchange via all_gather_object.
+    plists = [None] * world
+    d.all_gather_object(plists, projects)
+    projects = next(p for p in plists if p is not None)
+
+    buf = _pack(result, cell_order, projects)
+    use_cuda = d.get_backend() == "nccl"
+    t = torch.from_numpy(buf)
+    if use_cuda:
+        t = t.cuda()
+    d.all_reduce(t, op=d.ReduceOp.SUM)
+    buf = t.cpu().numpy()
+
+    return _unpack(buf, cell_order, projects)

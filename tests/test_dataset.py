@@ -1,0 +1,118 @@
+"""Unit tests for the collation / labeling / feature layer.
+
+Mirrors the coverage of the reference's test_experiment.py (its 4 pure
+functions) with fresh cases, plus the tests.json round trip.
+"""
+
+import io
+import json
+
+import numpy as np
+import pytest
+
+from flake16_framework_amd.constants import FLAKY, NON_FLAKY, OD_FLAKY
+from flake16_framework_amd.dataset.collate import (
+    update_collated_rusage, update_collated_runs,
+)
+from flake16_framework_amd.dataset.features import get_features_cov
+from flake16_framework_amd.dataset.labeling import get_req_runs_label
+from flake16_framework_amd.dataset.synthetic import make_synthetic_tests
+from flake16_framework_amd.dataset.tests_io import load_feat_lab_proj
+
+N_RUNS_SMALL = {"baseline": 4, "shuffle": 4, "testinspect": 1}
+
+
+def _runs(lines_per_run, mode):
+    collated_proj = [{}, None, None, None]
+    for run_n, lines in enumerate(lines_per_run):
+        update_collated_runs(io.StringIO("\n".join(lines)), mode, run_n,
+                             collated_proj)
+    return collated_proj
+
+
+class TestUpdateCollatedRuns:
+    def test_counts_and_min_runs(self):
+        collated = _runs([
+            ["passed\tt::a", "failed\tt::b"],
+            ["failed\tt::a", "failed\tt::b"],
+            ["passed\tt::a", "passed\tt::b"],
+        ], "baseline")
+        runs_a = collated[0]["t::a"][0]["baseline"]
+        runs_b = collated[0]["t::b"][0]["baseline"]
+        assert runs_a == [3, 1, 1, 0]
+        assert runs_b == [3, 2, 0, 2]
+
+    def test_outcome_substring_failed(self):
+        # the contract is substring matching: "xfailed" counts as failed
+        collated = _runs([["xfailed\tt::a"]], "shuffle")
+        assert collated[0]["t::a"][0]["shuffle"][1] == 1
+
+    def test_nodeid_with_tabs_preserved(self):
+        # only the FIRST tab splits outcome from nodeid
+        fd = io.StringIO("passed\tt::a[x\ty]")
+        collated_proj = [{}, None, None, None]
+        update_collated_runs(fd, "baseline", 0, collated_proj)
+        assert "t::a[x\ty]" in collated_proj[0]
+
+
+class TestLabeling:
+    @pytest.mark.parametrize("baseline,shuffle,expect", [
+        # incomplete counts -> dropped
+        ([3, 0, None, 0], [4, 0, None, 0], (0, None)),
+        ([4, 0, None, 0], [3, 0, None, 0], (0, None)),
+        # never fails anywhere -> non-flaky
+        ([4, 0, None, 0], [4, 0, None, 0], (0, NON_FLAKY)),
+        # baseline clean, shuffle fails -> OD, req = first failing shuffle run
+        ([4, 0, None, 0], [4, 2, 1, 0], (1, OD_FLAKY)),
+        # always fails everywhere -> non-flaky
+        ([4, 4, 0, None], [4, 4, 0, None], (0, NON_FLAKY)),
+        # always fails in baseline, shuffle passes once -> OD, req = first pass
+        ([4, 4, 0, None], [4, 3, 0, 2], (2, OD_FLAKY)),
+        # intermittent baseline -> NOD, req = max(first fail, first pass)
+        ([4, 2, 1, 0], [4, 0, None, 0], (1, FLAKY)),
+        ([4, 1, 3, 0], [4, 4, 0, None], (3, FLAKY)),
+    ])
+    def test_decision_table(self, baseline, shuffle, expect):
+        runs = {"baseline": baseline, "shuffle": shuffle}
+        assert get_req_runs_label(runs, N_RUNS_SMALL) == expect
+
+
+class TestCoverageFeatures:
+    def test_basic_counts(self):
+        cov = {"src/a.py": {1, 2, 3}, "tests/t.py": {5, 6}}
+        churn = {"src/a.py": {1: 4, 3: 2, 9: 7}}
+        n_lines, n_changes, n_src = get_features_cov(cov, {"tests/t.py"}, churn)
+        assert (n_lines, n_changes, n_src) == (5, 6, 3)
+
+    def test_empty(self):
+        assert get_features_cov({}, set(), {}) == (0, 0, 0)
+
+
+class TestRusage:
+    def test_parse(self):
+        fd = io.StringIO("1.5\t2\t3\t4\t5\t6.25\tt::a")
+        collated_proj = [{}, None, None, None]
+        update_collated_rusage(fd, collated_proj)
+        assert collated_proj[0]["t::a"][2] == [1.5, 2.0, 3.0, 4.0, 5.0, 6.25]
+
+
+class TestSyntheticAndLoad:
+    def test_shapes_and_determinism(self):
+        tests = make_synthetic_tests(n_tests=500, seed=3)
+        tests2 = make_synthetic_tests(n_tests=500, seed=3)
+        assert json.dumps(tests) == json.dumps(tests2)
+        total = sum(len(v) for v in tests.values())
+        assert total == 500
+
+        X, y, proj = load_feat_lab_proj(FLAKY, tuple(range(16)), tests=tests)
+        assert X.shape == (500, 16)
+        assert y.dtype == bool and 0 < y.sum() < 500
+        assert len(proj) == 500
+
+    def test_feature_subset(self):
+        tests = make_synthetic_tests(n_tests=200, seed=1)
+        X16, _, _ = load_feat_lab_proj(FLAKY, tuple(range(16)), tests=tests)
+        X7, _, _ = load_feat_lab_proj(FLAKY, (0, 1, 2, 3, 10, 11, 14),
+                                      tests=tests)
+        assert X7.shape[1] == 7
+        np.testing.assert_array_equal(X7[:, 3], X16[:, 3])
