@@ -39,7 +39,13 @@ def knn_indices(X_query, X_cand, k, skip_identity=False):
     excluded (use only when X_query IS X_cand)."""
     Xq = np.asarray(X_query, dtype=np.float64)
     Xc = np.asarray(X_cand, dtype=np.float64)
-    d2 = ((Xq[:, None, :] - Xc[None, :, :]) ** 2).sum(axis=2)
+    # Feature-sequential fp64 accumulation — the exact op order of the HIP
+    # knn kernel, so device and reference agree bitwise (tie behavior
+    # included).
+    d2 = np.zeros((Xq.shape[0], Xc.shape[0]))
+    for f in range(Xq.shape[1]):
+        diff = Xq[:, f, None] - Xc[None, :, f]
+        d2 = d2 + diff * diff
     if skip_identity:
         np.fill_diagonal(d2, np.inf)
     # stable argsort => ties broken by lower index

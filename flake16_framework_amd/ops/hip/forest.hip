@@ -1,0 +1,429 @@
+// Batched histogram-forest construction for MI355X (gfx950).
+//
+// One forest_fit call builds ALL trees of one grid cell (10 folds x
+// n_estimators jobs) device-resident: a level-synchronous work queue of
+// (job, node) items, one 256-thread workgroup per active node, per-node
+// LDS histograms (16 features x 256 bins x {total, class1} = 32 KiB),
+// split selection in fp64 (file compiled -ffp-contract=off so scores are
+// bit-identical to the numpy reference in models/forest_ref.py), and a
+// stable in-kernel partition into the next level's sample-index buffer.
+//
+// All randomness is Philox keyed on (tag, node sample-range, draw), so the
+// nondeterministic node-allocation order (atomics) cannot perturb trees.
+//
+// Reference semantics being implemented: sklearn 1.0.2 defaults for
+// DecisionTree/RandomForest/ExtraTrees (see models/forest_ref.py docstring;
+// reference experiment.py:96-98, 469, 473).
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+#include "philox.h"
+
+#define HBLK 256
+#define FPAD 16          // codes row stride (bytes); F <= 16
+#define LEAF_SENTINEL (-1)
+
+struct WorkItem {
+    int job;
+    int node;    // job-local node id
+    int start;   // job-local sidx range [start, end)
+    int end;
+    int depth;
+};
+
+struct ForestDev {
+    const uint8_t* __restrict__ codes;    // [R, FPAD]
+    const uint8_t* __restrict__ labels;   // [R]
+    const int* __restrict__ j_row_off;    // [J] fold data base row
+    const int* __restrict__ j_n;          // [J] samples per job
+    const long* __restrict__ j_sidx_off;  // [J] base into sidx buffers
+    const long* __restrict__ j_node_off;  // [J] base into node arrays
+    const int* __restrict__ j_key;        // [J] philox k1 per job
+    int* __restrict__ node_alloc;         // [J]
+    int* __restrict__ nfeat;              // per node (LEAF_SENTINEL = leaf)
+    int* __restrict__ nsplit;
+    int* __restrict__ nleft;              // left child id; right = left + 1
+    float* __restrict__ ncnt0;
+    float* __restrict__ ncnt1;
+    const int* __restrict__ sidx_cur;
+    int* __restrict__ sidx_nxt;
+    const WorkItem* __restrict__ cur;
+    const int* __restrict__ cur_count;
+    WorkItem* __restrict__ nxt;
+    int* __restrict__ nxt_count;
+    int* __restrict__ err_flag;
+    int F;
+    int max_features;
+    int splitter_random;
+    uint32_t seed;
+    int work_cap;
+};
+
+// ---------------------------------------------------------------------------
+// Init: fill per-job sample indices (bootstrap or identity) and root items.
+// Bootstrap draw i: bounded(philox(TAG_BOOTSTRAP, 0, 0, i), n)  — matches
+// forest_ref.fit_forest.
+// ---------------------------------------------------------------------------
+__global__ void forest_init_kernel(
+    const int* __restrict__ j_row_off, const int* __restrict__ j_n,
+    const long* __restrict__ j_sidx_off, const int* __restrict__ j_key,
+    int* __restrict__ node_alloc, int* __restrict__ sidx,
+    WorkItem* __restrict__ work, int bootstrap, uint32_t seed) {
+    int job = blockIdx.x;
+    int n = j_n[job];
+    long off = j_sidx_off[job];
+    int row0 = j_row_off[job];
+    uint32_t key = (uint32_t)j_key[job];
+
+    for (int i = threadIdx.x; i < n; i += blockDim.x) {
+        int s;
+        if (bootstrap) {
+            uint32_t u = philox_draw(TAG_BOOTSTRAP, 0u, 0u, (uint32_t)i,
+                                     seed, key);
+            s = (int)philox_bounded(u, (uint32_t)n);
+        } else {
+            s = i;
+        }
+        sidx[off + i] = row0 + s;
+    }
+    if (threadIdx.x == 0) {
+        node_alloc[job] = 1;
+        work[job] = {job, 0, 0, n, 0};
+    }
+}
+
+// ---------------------------------------------------------------------------
+// The level kernel: histogram + split + partition, one workgroup per item.
+// ---------------------------------------------------------------------------
+__launch_bounds__(HBLK)
+__global__ void hist_split_kernel(ForestDev a) {
+    __shared__ uint32_t hist_n[FPAD * 256];
+    __shared__ uint32_t hist_1[FPAD * 256];
+    __shared__ int sh_scan[HBLK];
+    __shared__ int sh_bmin[FPAD], sh_bmax[FPAD];
+    __shared__ int sh_cand[FPAD], sh_ncand;
+    __shared__ double sh_score[FPAD];
+    __shared__ int sh_bin[FPAD], sh_nL[FPAD];
+    __shared__ int sh_bestf, sh_bestbin, sh_bestnL;
+    __shared__ int sh_loff, sh_roff;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int n_items = *a.cur_count;
+
+    for (int wi = blockIdx.x; wi < n_items; wi += gridDim.x) {
+        WorkItem it = a.cur[wi];
+        const int n = it.end - it.start;
+        const long sbase = a.j_sidx_off[it.job];
+        const long nbase = a.j_node_off[it.job];
+        const uint32_t key = (uint32_t)a.j_key[it.job];
+        const int F = a.F;
+
+        // Phase 0: zero LDS histograms.
+        for (int i = tid; i < F * 256; i += HBLK) {
+            hist_n[i] = 0;
+            hist_1[i] = 0;
+        }
+        __syncthreads();
+
+        // Phase 1: accumulate. One uint4 = the sample's 16 packed bin codes.
+        for (int i = it.start + tid; i < it.end; i += HBLK) {
+            int row = a.sidx_cur[sbase + i];
+            uint4 cw = *reinterpret_cast<const uint4*>(
+                a.codes + (size_t)row * FPAD);
+            uint32_t w[4] = {cw.x, cw.y, cw.z, cw.w};
+            int lab = a.labels[row];
+            for (int f = 0; f < F; ++f) {
+                uint32_t b = (w[f >> 2] >> ((f & 3) * 8)) & 0xFFu;
+                atomicAdd(&hist_n[f * 256 + b], 1u);
+                if (lab) atomicAdd(&hist_1[f * 256 + b], 1u);
+            }
+        }
+        __syncthreads();
+
+        // Phase 2: class counts from feature-0 histogram.
+        int v = (int)hist_1[tid];
+        sh_scan[tid] = v;
+        __syncthreads();
+        for (int d = HBLK / 2; d > 0; d >>= 1) {
+            if (tid < d) sh_scan[tid] += sh_scan[tid + d];
+            __syncthreads();
+        }
+        const int c1 = sh_scan[0];
+        const int c0 = n - c1;
+        __syncthreads();
+
+        if (tid == 0) {
+            a.ncnt0[nbase + it.node] = (float)c0;
+            a.ncnt1[nbase + it.node] = (float)c1;
+        }
+
+        if (n < 2 || c0 == 0 || c1 == 0) {
+            __syncthreads();
+            continue;  // leaf (nfeat stays LEAF_SENTINEL)
+        }
+
+        // Phase 3: occupied-bin range per feature.
+        if (tid < F) {
+            int bmin = -1, bmax = -1;
+            for (int b = 0; b < 256; ++b)
+                if (hist_n[tid * 256 + b]) { bmin = b; break; }
+            for (int b = 255; b >= 0; --b)
+                if (hist_n[tid * 256 + b]) { bmax = b; break; }
+            sh_bmin[tid] = bmin;
+            sh_bmax[tid] = bmax;
+        }
+        __syncthreads();
+
+        // Phase 4 (thread 0): feature permutation (partial Fisher-Yates,
+        // counters (TAG_FEATSEL|depth<<8, start, end, i)) and the candidate
+        // list: walk perm, non-constant features until max_features.
+        if (tid == 0) {
+            int perm[FPAD];
+            for (int f = 0; f < F; ++f) perm[f] = f;
+            uint32_t tag = TAG_FEATSEL | ((uint32_t)(it.depth & 0xFF) << 8);
+            for (int i = 0; i < F - 1; ++i) {
+                uint32_t u = philox_draw(tag, (uint32_t)it.start,
+                                         (uint32_t)it.end, (uint32_t)i,
+                                         a.seed, key);
+                int j = i + (int)philox_bounded(u, (uint32_t)(F - i));
+                int t = perm[i]; perm[i] = perm[j]; perm[j] = t;
+            }
+            int nc = 0;
+            for (int i = 0; i < F && nc < a.max_features; ++i) {
+                int f = perm[i];
+                if (sh_bmin[f] != sh_bmax[f]) sh_cand[nc++] = f;
+            }
+            sh_ncand = nc;
+        }
+        __syncthreads();
+
+        const int ncand = sh_ncand;
+        if (ncand == 0) {
+            __syncthreads();
+            continue;  // all features constant: leaf
+        }
+
+        // Phase 5: evaluate candidates; wave w handles candidates w, w+4, ...
+        // Scores in fp64, expression order matching the reference
+        // (-ffp-contract=off).
+        for (int ci = wave; ci < ncand; ci += HBLK / 64) {
+            const int f = sh_cand[ci];
+            const int bmin = sh_bmin[f], bmax = sh_bmax[f];
+
+            // Per-lane 4-bin partial sums, then wave-inclusive scan.
+            int ln[4], l1[4];
+            int tn = 0, t1 = 0;
+            for (int k = 0; k < 4; ++k) {
+                int b = lane * 4 + k;
+                ln[k] = (int)hist_n[f * 256 + b];
+                l1[k] = (int)hist_1[f * 256 + b];
+                tn += ln[k];
+                t1 += l1[k];
+            }
+            int scn = tn, sc1 = t1;
+            for (int d = 1; d < 64; d <<= 1) {
+                int un = __shfl_up(scn, d);
+                int u1 = __shfl_up(sc1, d);
+                if (lane >= d) { scn += un; sc1 += u1; }
+            }
+            const int excl_n = scn - tn, excl_1 = sc1 - t1;
+
+            double best_s = -1.0;  // all real scores are > 0
+            int best_b = -1, best_nl = 0;
+
+            if (a.splitter_random) {
+                uint32_t tag = TAG_THRESH | ((uint32_t)(it.depth & 0xFF) << 8);
+                uint32_t u = philox_draw(tag, (uint32_t)it.start,
+                                         (uint32_t)it.end, (uint32_t)f,
+                                         a.seed, key);
+                int b = bmin + (int)philox_bounded(u, (uint32_t)(bmax - bmin));
+                // the lane owning bin b evaluates it
+                if (b >= lane * 4 && b < lane * 4 + 4) {
+                    int cn = excl_n, c1f = excl_1;
+                    for (int k = 0; k <= b - lane * 4; ++k) {
+                        cn += ln[k];
+                        c1f += l1[k];
+                    }
+                    long nL = cn, n1L = c1f;
+                    long n0L = nL - n1L, nR = n - nL;
+                    long n1R = c1 - n1L, n0R = c0 - n0L;
+                    best_s = (double)(n0L * n0L + n1L * n1L) / (double)nL
+                           + (double)(n0R * n0R + n1R * n1R) / (double)nR;
+                    best_b = b;
+                    best_nl = (int)nL;
+                }
+            } else {
+                int cn = excl_n, c1f = excl_1;
+                for (int k = 0; k < 4; ++k) {
+                    int b = lane * 4 + k;
+                    cn += ln[k];
+                    c1f += l1[k];
+                    if (b < bmin || b >= bmax) continue;
+                    long nL = cn, n1L = c1f;
+                    long n0L = nL - n1L, nR = n - nL;
+                    long n1R = c1 - n1L, n0R = c0 - n0L;
+                    double s = (double)(n0L * n0L + n1L * n1L) / (double)nL
+                             + (double)(n0R * n0R + n1R * n1R) / (double)nR;
+                    if (s > best_s) { best_s = s; best_b = b; best_nl = (int)nL; }
+                }
+            }
+
+            // Wave arg-max with first-of-ties (lowest bin) tie-break.
+            for (int d = 32; d > 0; d >>= 1) {
+                double os = __shfl_down(best_s, d);
+                int ob = __shfl_down(best_b, d);
+                int onl = __shfl_down(best_nl, d);
+                if (os > best_s || (os == best_s && ob != -1 &&
+                                    (best_b == -1 || ob < best_b))) {
+                    best_s = os; best_b = ob; best_nl = onl;
+                }
+            }
+            if (lane == 0) {
+                sh_score[ci] = best_s;
+                sh_bin[ci] = best_b;
+                sh_nL[ci] = best_nl;
+            }
+        }
+        __syncthreads();
+
+        // Phase 6 (thread 0): sequential select in perm order, strict >.
+        if (tid == 0) {
+            double best_s = -1.0e300;
+            int bf = -1, bb = -1, bnl = 0;
+            for (int ci = 0; ci < ncand; ++ci) {
+                if (sh_bin[ci] >= 0 && sh_score[ci] > best_s) {
+                    best_s = sh_score[ci];
+                    bf = sh_cand[ci];
+                    bb = sh_bin[ci];
+                    bnl = sh_nL[ci];
+                }
+            }
+            sh_bestf = bf;
+            sh_bestbin = bb;
+            sh_bestnL = bnl;
+            if (bf >= 0) {
+                int l = atomicAdd(&a.node_alloc[it.job], 2);
+                a.nfeat[nbase + it.node] = bf;
+                a.nsplit[nbase + it.node] = bb;
+                a.nleft[nbase + it.node] = l;
+                int qi = atomicAdd(a.nxt_count, 2);
+                if (qi + 1 < a.work_cap) {
+                    a.nxt[qi] = {it.job, l, it.start, it.start + bnl,
+                                 it.depth + 1};
+                    a.nxt[qi + 1] = {it.job, l + 1, it.start + bnl, it.end,
+                                     it.depth + 1};
+                } else {
+                    atomicExch(a.err_flag, 1);
+                }
+            }
+            sh_loff = 0;
+            sh_roff = 0;
+        }
+        __syncthreads();
+
+        const int bf = sh_bestf;
+        if (bf < 0) {
+            __syncthreads();
+            continue;  // no valid split: leaf
+        }
+        const int bb = sh_bestbin;
+        const int nL = sh_bestnL;
+
+        // Phase 7: stable partition into sidx_nxt, 256-wide tiles.
+        for (int base = it.start; base < it.end; base += HBLK) {
+            const int i = base + tid;
+            const bool valid = i < it.end;
+            int row = 0, flag = 0;
+            if (valid) {
+                row = a.sidx_cur[sbase + i];
+                uint32_t b = a.codes[(size_t)row * FPAD + bf];
+                flag = (int)(b <= (uint32_t)bb);
+            }
+            sh_scan[tid] = flag;
+            __syncthreads();
+            for (int d = 1; d < HBLK; d <<= 1) {
+                int t = (tid >= d) ? sh_scan[tid - d] : 0;
+                __syncthreads();
+                sh_scan[tid] += t;
+                __syncthreads();
+            }
+            const int incl = sh_scan[tid];
+            const int tile_left = sh_scan[HBLK - 1];
+            const int tile_n = min(HBLK, it.end - base);
+            if (valid) {
+                const int excl = incl - flag;
+                if (flag)
+                    a.sidx_nxt[sbase + it.start + sh_loff + excl] = row;
+                else
+                    a.sidx_nxt[sbase + it.start + nL + sh_roff
+                               + (i - base) - excl] = row;
+            }
+            __syncthreads();
+            if (tid == 0) {
+                sh_loff += tile_left;
+                sh_roff += tile_n - tile_left;
+            }
+            __syncthreads();
+        }
+        __syncthreads();
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Ensemble prediction + confusion.
+//
+// One thread per (fold, test-row) pair; trees visited in job order so the
+// fp64 probability accumulation matches forest_ref.predict_forest exactly.
+// Confusion counts go to per-(project, k) atomics, k = 2*y + pred - 1 with
+// true negatives skipped (reference experiment.py:476-483).
+// ---------------------------------------------------------------------------
+__global__ void predict_confusion_kernel(
+    const uint8_t* __restrict__ codes_test,   // [M, FPAD] full-dataset codes
+    const uint8_t* __restrict__ y_test,       // [M]
+    const int* __restrict__ proj_id,          // [M] project index per row
+    const int* __restrict__ pair_row,         // [P] dataset row of pair
+    const int* __restrict__ pair_fold,        // [P] fold of pair
+    int n_pairs,
+    const long* __restrict__ j_node_off,      // [J]
+    const int* __restrict__ nfeat, const int* __restrict__ nsplit,
+    const int* __restrict__ nleft,
+    const float* __restrict__ ncnt0, const float* __restrict__ ncnt1,
+    int trees_per_fold,
+    uint8_t* __restrict__ pred_out,           // [P]
+    int* __restrict__ confusion,              // [n_proj+1, 3]
+    int n_proj) {
+    int p = blockIdx.x * blockDim.x + threadIdx.x;
+    if (p >= n_pairs) return;
+
+    const int row = pair_row[p];
+    const int fold = pair_fold[p];
+    const uint8_t* cr = codes_test + (size_t)row * FPAD;
+
+    double acc0 = 0.0, acc1 = 0.0;
+    for (int t = 0; t < trees_per_fold; ++t) {
+        const long nbase = j_node_off[fold * trees_per_fold + t];
+        int node = 0;
+        int f = nfeat[nbase];
+        while (f != LEAF_SENTINEL) {
+            int go_right = (int)cr[f] > nsplit[nbase + node];
+            node = nleft[nbase + node] + go_right;
+            f = nfeat[nbase + node];
+        }
+        double c0 = (double)ncnt0[nbase + node];
+        double c1 = (double)ncnt1[nbase + node];
+        double tot = c0 + c1;
+        acc0 += c0 / tot;
+        acc1 += c1 / tot;
+    }
+    const int pred = acc1 > acc0;
+    pred_out[p] = (uint8_t)pred;
+
+    const int k = 2 * (int)y_test[row] + pred - 1;
+    if (k >= 0) {
+        atomicAdd(&confusion[proj_id[row] * 3 + k], 1);
+        atomicAdd(&confusion[n_proj * 3 + k], 1);
+    }
+}

@@ -1,0 +1,307 @@
+// Torch-extension host layer for the flake16 MI355X kernels.
+//
+// Single translation unit: includes the kernel files and exposes
+// tensor-based entry points.  Compiled by hipcc for gfx950 with
+// -ffp-contract=off (bit-parity of fp64 split scores / fp32 SMOTE
+// interpolation with the numpy reference).
+
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include <c10/cuda/CUDAGuard.h>
+
+#include "forest.hip"
+#include "knn_balance.hip"
+#include "scaler_pca.hip"
+
+#include <vector>
+
+#define CHECK_HIP(expr)                                                     \
+    do {                                                                    \
+        hipError_t _e = (expr);                                             \
+        TORCH_CHECK(_e == hipSuccess, "HIP error: ",                        \
+                    hipGetErrorString(_e));                                 \
+    } while (0)
+
+static hipStream_t current_stream() {
+    return (hipStream_t)at::cuda::getCurrentCUDAStream().stream();
+}
+
+// ---------------------------------------------------------------------------
+// forest_fit
+// ---------------------------------------------------------------------------
+std::vector<at::Tensor> forest_fit(
+    at::Tensor codes, at::Tensor labels, at::Tensor j_row_off,
+    at::Tensor j_n, at::Tensor j_key, int64_t F, int64_t max_features,
+    bool bootstrap, bool splitter_random, int64_t seed) {
+    TORCH_CHECK(codes.is_cuda() && codes.dtype() == at::kByte &&
+                codes.size(1) == FPAD && codes.is_contiguous());
+    TORCH_CHECK(labels.is_cuda() && labels.dtype() == at::kByte);
+    const at::cuda::OptionalCUDAGuard guard(codes.device());
+
+    const int J = j_n.size(0);
+    auto opts_i32 = codes.options().dtype(at::kInt);
+    auto opts_i64 = codes.options().dtype(at::kLong);
+    auto opts_f32 = codes.options().dtype(at::kFloat);
+
+    auto j_n_cpu = j_n.to(at::kCPU);
+    const int* jn = j_n_cpu.data_ptr<int>();
+    std::vector<long> sidx_off(J), node_off(J);
+    long S = 0, Ntot = 0;
+    for (int j = 0; j < J; ++j) {
+        sidx_off[j] = S;
+        node_off[j] = Ntot;
+        S += jn[j];
+        Ntot += 2L * jn[j] + 1;
+    }
+    TORCH_CHECK(S > 0, "empty forest_fit batch");
+
+    auto j_sidx_off = at::from_blob(sidx_off.data(), {J}, at::kLong)
+                          .to(codes.device());
+    auto j_node_off = at::from_blob(node_off.data(), {J}, at::kLong)
+                          .to(codes.device());
+
+    auto nfeat = at::full({Ntot}, LEAF_SENTINEL, opts_i32);
+    auto nsplit = at::zeros({Ntot}, opts_i32);
+    auto nleft = at::zeros({Ntot}, opts_i32);
+    auto ncnt0 = at::zeros({Ntot}, opts_f32);
+    auto ncnt1 = at::zeros({Ntot}, opts_f32);
+    auto node_alloc = at::zeros({J}, opts_i32);
+
+    auto sidx_a = at::empty({S}, opts_i32);
+    auto sidx_b = at::empty({S}, opts_i32);
+    const long work_cap = S + J + 2;
+    auto work_a = at::empty({work_cap * (long)sizeof(WorkItem)},
+                            codes.options().dtype(at::kByte));
+    auto work_b = at::empty({work_cap * (long)sizeof(WorkItem)},
+                            codes.options().dtype(at::kByte));
+    auto counts = at::zeros({2}, opts_i32);
+    auto err = at::zeros({1}, opts_i32);
+
+    hipStream_t stream = current_stream();
+
+    forest_init_kernel<<<J, HBLK, 0, stream>>>(
+        j_row_off.data_ptr<int>(), j_n.data_ptr<int>(),
+        j_sidx_off.data_ptr<long>(), j_key.data_ptr<int>(),
+        node_alloc.data_ptr<int>(), sidx_a.data_ptr<int>(),
+        (WorkItem*)work_a.data_ptr(), bootstrap ? 1 : 0, (uint32_t)seed);
+    counts.narrow(0, 0, 1).fill_((int)J);
+
+    const int PINSZ = 64;
+    auto pinned = at::empty({PINSZ}, at::TensorOptions()
+                                         .dtype(at::kInt)
+                                         .pinned_memory(true));
+    int* pinned_p = pinned.data_ptr<int>();
+
+    ForestDev a{};
+    a.codes = codes.data_ptr<uint8_t>();
+    a.labels = labels.data_ptr<uint8_t>();
+    a.j_row_off = j_row_off.data_ptr<int>();
+    a.j_n = j_n.data_ptr<int>();
+    a.j_sidx_off = j_sidx_off.data_ptr<long>();
+    a.j_node_off = j_node_off.data_ptr<long>();
+    a.j_key = j_key.data_ptr<int>();
+    a.node_alloc = node_alloc.data_ptr<int>();
+    a.nfeat = nfeat.data_ptr<int>();
+    a.nsplit = nsplit.data_ptr<int>();
+    a.nleft = nleft.data_ptr<int>();
+    a.ncnt0 = ncnt0.data_ptr<float>();
+    a.ncnt1 = ncnt1.data_ptr<float>();
+    a.err_flag = err.data_ptr<int>();
+    a.F = (int)F;
+    a.max_features = (int)max_features;
+    a.splitter_random = splitter_random ? 1 : 0;
+    a.seed = (uint32_t)seed;
+    a.work_cap = (int)work_cap;
+
+    const int GRID = 4096;
+    const int CHUNK = 8;
+    int cur = 0;
+    long lev = 0;
+    bool done = false;
+    while (!done) {
+        for (int c = 0; c < CHUNK; ++c, ++lev) {
+            int nx = cur ^ 1;
+            CHECK_HIP(hipMemsetAsync(counts.data_ptr<int>() + nx, 0, 4,
+                                     stream));
+            a.sidx_cur = (cur == 0 ? sidx_a : sidx_b).data_ptr<int>();
+            a.sidx_nxt = (cur == 0 ? sidx_b : sidx_a).data_ptr<int>();
+            a.cur = (const WorkItem*)(cur == 0 ? work_a : work_b).data_ptr();
+            a.nxt = (WorkItem*)(cur == 0 ? work_b : work_a).data_ptr();
+            a.cur_count = counts.data_ptr<int>() + cur;
+            a.nxt_count = counts.data_ptr<int>() + nx;
+            hist_split_kernel<<<GRID, HBLK, 0, stream>>>(a);
+            CHECK_HIP(hipMemcpyAsync(pinned_p + (c % PINSZ),
+                                     counts.data_ptr<int>() + nx, 4,
+                                     hipMemcpyDeviceToHost, stream));
+            cur = nx;
+        }
+        CHECK_HIP(hipStreamSynchronize(stream));
+        for (int c = 0; c < CHUNK; ++c)
+            if (pinned_p[c] == 0) { done = true; break; }
+        TORCH_CHECK(lev < 8192, "forest_fit: depth limit exceeded");
+    }
+
+    TORCH_CHECK(err.to(at::kCPU).item<int>() == 0,
+                "forest_fit: work-queue capacity exceeded");
+
+    return {nfeat, nsplit, nleft, ncnt0, ncnt1, j_node_off, node_alloc};
+}
+
+// ---------------------------------------------------------------------------
+// forest_predict_confusion
+// ---------------------------------------------------------------------------
+std::vector<at::Tensor> forest_predict_confusion(
+    at::Tensor codes_test, at::Tensor y_test, at::Tensor proj_id,
+    at::Tensor pair_row, at::Tensor pair_fold, at::Tensor j_node_off,
+    at::Tensor nfeat, at::Tensor nsplit, at::Tensor nleft,
+    at::Tensor ncnt0, at::Tensor ncnt1, int64_t trees_per_fold,
+    int64_t n_proj) {
+    const at::cuda::OptionalCUDAGuard guard(codes_test.device());
+    const int P = pair_row.size(0);
+    auto pred = at::zeros({P}, codes_test.options().dtype(at::kByte));
+    auto confusion = at::zeros({n_proj + 1, 3},
+                               codes_test.options().dtype(at::kInt));
+    const int grid = (P + 255) / 256;
+    predict_confusion_kernel<<<grid, 256, 0, current_stream()>>>(
+        codes_test.data_ptr<uint8_t>(), y_test.data_ptr<uint8_t>(),
+        proj_id.data_ptr<int>(), pair_row.data_ptr<int>(),
+        pair_fold.data_ptr<int>(), P, j_node_off.data_ptr<long>(),
+        nfeat.data_ptr<int>(), nsplit.data_ptr<int>(),
+        nleft.data_ptr<int>(), ncnt0.data_ptr<float>(),
+        ncnt1.data_ptr<float>(), (int)trees_per_fold,
+        pred.data_ptr<uint8_t>(), confusion.data_ptr<int>(), (int)n_proj);
+    return {pred, confusion};
+}
+
+// ---------------------------------------------------------------------------
+// knn / balancing / binning
+// ---------------------------------------------------------------------------
+at::Tensor knn(at::Tensor X, int64_t k, bool skip_identity) {
+    const at::cuda::OptionalCUDAGuard guard(X.device());
+    TORCH_CHECK(X.is_cuda() && X.dtype() == at::kFloat &&
+                X.size(1) == FPAD && X.is_contiguous());
+    TORCH_CHECK(k >= 1 && k <= KMAX);
+    const int n = X.size(0);
+    auto out = at::empty({n, k}, X.options().dtype(at::kInt));
+    const int grid = (n + KNN_BLK - 1) / KNN_BLK;
+    knn_kernel<<<grid, KNN_BLK, 0, current_stream()>>>(
+        X.data_ptr<float>(), n, (int)k, skip_identity ? 1 : 0,
+        out.data_ptr<int>());
+    return out;
+}
+
+at::Tensor smote_interpolate(at::Tensor X, at::Tensor min_rows,
+                             at::Tensor nn, int64_t n_new, int64_t k0,
+                             int64_t k1) {
+    const at::cuda::OptionalCUDAGuard guard(X.device());
+    const int n_min = min_rows.size(0);
+    const int k = nn.size(1);
+    auto out = at::empty({n_new, FPAD}, X.options());
+    const int grid = (n_new + 255) / 256;
+    smote_kernel<<<grid, 256, 0, current_stream()>>>(
+        X.data_ptr<float>(), min_rows.data_ptr<int>(), nn.data_ptr<int>(),
+        n_min, k, (int)n_new, (uint32_t)k0, (uint32_t)k1,
+        out.data_ptr<float>());
+    return out;
+}
+
+at::Tensor enn_keep(at::Tensor y, at::Tensor nn, int64_t n_neighbors,
+                    int64_t maj_label, bool clean_all) {
+    const at::cuda::OptionalCUDAGuard guard(y.device());
+    const int n = y.size(0);
+    auto keep = at::empty({n}, y.options());
+    const int grid = (n + 255) / 256;
+    enn_keep_kernel<<<grid, 256, 0, current_stream()>>>(
+        y.data_ptr<uint8_t>(), nn.data_ptr<int>(), n, (int)nn.size(1),
+        (int)n_neighbors, (int)maj_label, clean_all ? 1 : 0,
+        keep.data_ptr<uint8_t>());
+    return keep;
+}
+
+at::Tensor tomek_keep(at::Tensor y, at::Tensor nn1, int64_t maj_label,
+                      bool remove_all) {
+    const at::cuda::OptionalCUDAGuard guard(y.device());
+    const int n = y.size(0);
+    auto keep = at::empty({n}, y.options());
+    const int grid = (n + 255) / 256;
+    tomek_keep_kernel<<<grid, 256, 0, current_stream()>>>(
+        y.data_ptr<uint8_t>(), nn1.data_ptr<int>(), n, (int)maj_label,
+        remove_all ? 1 : 0, keep.data_ptr<uint8_t>());
+    return keep;
+}
+
+at::Tensor bin_codes_dev(at::Tensor X, at::Tensor cuts, at::Tensor cut_off,
+                         int64_t F) {
+    const at::cuda::OptionalCUDAGuard guard(X.device());
+    const int n = X.size(0);
+    auto codes = at::zeros({n, FPAD}, X.options().dtype(at::kByte));
+    const int grid = (n + 255) / 256;
+    bin_codes_kernel<<<grid, 256, 0, current_stream()>>>(
+        X.data_ptr<float>(), cuts.data_ptr<float>(), cut_off.data_ptr<int>(),
+        n, (int)F, codes.data_ptr<uint8_t>());
+    return codes;
+}
+
+// ---------------------------------------------------------------------------
+// scaler / pca  (fp64 in, fp64 out)
+// ---------------------------------------------------------------------------
+at::Tensor scaler_fit_transform_dev(at::Tensor X) {
+    const at::cuda::OptionalCUDAGuard guard(X.device());
+    TORCH_CHECK(X.dtype() == at::kDouble && X.size(1) == FPAD);
+    const int n = X.size(0);
+    auto mean = at::empty({FPAD}, X.options());
+    auto scale = at::empty({FPAD}, X.options());
+    auto out = at::empty_like(X);
+    hipStream_t s = current_stream();
+    col_mean_kernel<<<FPAD, RBLK, 0, s>>>(X.data_ptr<double>(), n,
+                                          mean.data_ptr<double>());
+    col_scale_kernel<<<FPAD, RBLK, 0, s>>>(X.data_ptr<double>(), n,
+                                           mean.data_ptr<double>(),
+                                           scale.data_ptr<double>());
+    scale_transform_kernel<<<(n + 255) / 256, 256, 0, s>>>(
+        X.data_ptr<double>(), n, mean.data_ptr<double>(),
+        scale.data_ptr<double>(), out.data_ptr<double>());
+    return out;
+}
+
+at::Tensor pca_fit_transform_dev(at::Tensor X, int64_t F) {
+    const at::cuda::OptionalCUDAGuard guard(X.device());
+    TORCH_CHECK(X.dtype() == at::kDouble && X.size(1) == FPAD);
+    const int n = X.size(0);
+    auto mean = at::empty({FPAD}, X.options());
+    auto C = at::zeros({FPAD, FPAD}, X.options());
+    auto V = at::zeros({FPAD, FPAD}, X.options());
+    auto evals = at::zeros({FPAD}, X.options());
+    auto T = at::empty_like(X);
+    hipStream_t s = current_stream();
+    col_mean_kernel<<<FPAD, RBLK, 0, s>>>(X.data_ptr<double>(), n,
+                                          mean.data_ptr<double>());
+    gram_kernel<<<FPAD * FPAD, RBLK, 0, s>>>(X.data_ptr<double>(), n,
+                                             mean.data_ptr<double>(),
+                                             C.data_ptr<double>());
+    jacobi_eigen_kernel<<<1, 1, 0, s>>>(C.data_ptr<double>(),
+                                        V.data_ptr<double>(),
+                                        evals.data_ptr<double>(), (int)F);
+    pca_project_kernel<<<(n + 255) / 256, 256, 0, s>>>(
+        X.data_ptr<double>(), n, mean.data_ptr<double>(),
+        V.data_ptr<double>(), (int)F, T.data_ptr<double>());
+    pca_signflip_kernel<<<FPAD, RBLK, 0, s>>>(T.data_ptr<double>(), n,
+                                              (int)F);
+    return T;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("forest_fit", &forest_fit,
+          "Batched histogram-forest fit (gfx950)");
+    m.def("forest_predict_confusion", &forest_predict_confusion,
+          "Ensemble predict + confusion accumulation");
+    m.def("knn", &knn, "Brute-force k-NN (fp64 distances)");
+    m.def("smote_interpolate", &smote_interpolate, "SMOTE synthesis");
+    m.def("enn_keep", &enn_keep, "ENN keep-mask");
+    m.def("tomek_keep", &tomek_keep, "Tomek-link keep-mask");
+    m.def("bin_codes", &bin_codes_dev, "Quantile-bin codes");
+    m.def("scaler_fit_transform", &scaler_fit_transform_dev,
+          "StandardScaler fit_transform");
+    m.def("pca_fit_transform", &pca_fit_transform_dev,
+          "Full PCA fit_transform");
+}
