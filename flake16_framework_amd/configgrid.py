@@ -78,6 +78,20 @@ def resolve(config_keys):
     return [CONFIG_GRID[i][k] for i, k in enumerate(config_keys)]
 
 
+def balance_group_index(config_keys):
+    """Index of the cell's (flaky-type, feature-set, preprocessing,
+    balancing) combination in product order — 72 groups.
+
+    The balanced training folds are identical for the 3 cells of a balance
+    group (the model axis does not affect balancing), so balancing RNG is
+    keyed on the GROUP id and the engine computes each group's balanced
+    folds once (reference recomputes per cell; sharing changes no output).
+    """
+    import itertools as _it
+    combos = list(_it.product(*[d.keys() for d in CONFIG_GRID[:4]]))
+    return combos.index(tuple(config_keys[:4]))
+
+
 def cell_cost_estimate(config_keys):
     """Rough relative cost of a cell, for load balancing across ranks.
 

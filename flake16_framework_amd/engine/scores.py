@@ -35,11 +35,19 @@ GLOBAL_SEED = 0
 N_FOLDS = 10
 
 
-def job_ids_for(cell_idx, fold):
+def job_ids_for(config_keys, cell_idx, fold):
     """Philox key-ids for one (cell, fold): (balance_k1, tree_job_base).
-    Tree t of this fold uses k1 = tree_job_base + t (t < 128)."""
+
+    Balancing is keyed on the cell's BALANCE GROUP (the 72 distinct
+    (flaky, feature-set, preproc, balancing) combos) so the engine can share
+    balanced folds across the 3 model-axis cells of a group; trees are keyed
+    per cell.  Tree t of this fold uses k1 = tree_job_base + t (t < 128).
+    Philox tags domain-separate the two key spaces.
+    """
+    from ..configgrid import balance_group_index
+    bal_k1 = balance_group_index(config_keys) * N_FOLDS + fold
     ctx = cell_idx * N_FOLDS + fold
-    return ctx, ctx * 128
+    return bal_k1, ctx * 128
 
 
 def evaluate_cell_ref(config_keys, cell_idx, tests=None, tests_file=None,
@@ -68,7 +76,7 @@ def evaluate_cell_ref(config_keys, cell_idx, tests=None, tests_file=None,
     for i, (train, test) in enumerate(
             stratified_kfold_split(labels, n_splits=N_FOLDS,
                                    random_state=seed)):
-        bal_k1, job_base = job_ids_for(cell_idx, i)
+        bal_k1, job_base = job_ids_for(config_keys, cell_idx, i)
         Xb, yb = apply_balancing(X[train], labels[train], balancing,
                                  seed, bal_k1)
         codes_tr = bin_codes(Xb, cuts)
@@ -107,14 +115,16 @@ def run_scores(tests_file=None, tests=None, backend="auto", cells=None,
     if backend == "auto":
         backend = _auto_backend()
 
+    context = None
+    if backend == "hip":
+        from .hip_cell import SweepContext
+        context = SweepContext(tests=tests, tests_file=tests_file, seed=seed)
+
     out = {}
     t_start = time.time()
     for n_done, (cell_idx, config_keys) in enumerate(all_cells):
         if backend == "hip":
-            from .hip_cell import evaluate_cell_hip
-            out[config_keys] = evaluate_cell_hip(
-                config_keys, cell_idx, tests=tests, tests_file=tests_file,
-                seed=seed)
+            out[config_keys] = context.evaluate_cell(config_keys, cell_idx)
         else:
             out[config_keys] = evaluate_cell_ref(
                 config_keys, cell_idx, tests=tests, tests_file=tests_file,

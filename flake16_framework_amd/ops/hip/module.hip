@@ -291,17 +291,28 @@ at::Tensor pca_fit_transform_dev(at::Tensor X, int64_t F) {
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    // gil_scoped_release: the forest_fit level loop blocks on stream syncs;
+    // releasing the GIL lets other Python threads drive their own streams.
     m.def("forest_fit", &forest_fit,
+          py::call_guard<py::gil_scoped_release>(),
           "Batched histogram-forest fit (gfx950)");
     m.def("forest_predict_confusion", &forest_predict_confusion,
+          py::call_guard<py::gil_scoped_release>(),
           "Ensemble predict + confusion accumulation");
-    m.def("knn", &knn, "Brute-force k-NN (fp64 distances)");
-    m.def("smote_interpolate", &smote_interpolate, "SMOTE synthesis");
-    m.def("enn_keep", &enn_keep, "ENN keep-mask");
-    m.def("tomek_keep", &tomek_keep, "Tomek-link keep-mask");
-    m.def("bin_codes", &bin_codes_dev, "Quantile-bin codes");
+    m.def("knn", &knn, py::call_guard<py::gil_scoped_release>(),
+          "Brute-force k-NN (fp64 distances)");
+    m.def("smote_interpolate", &smote_interpolate,
+          py::call_guard<py::gil_scoped_release>(), "SMOTE synthesis");
+    m.def("enn_keep", &enn_keep, py::call_guard<py::gil_scoped_release>(),
+          "ENN keep-mask");
+    m.def("tomek_keep", &tomek_keep,
+          py::call_guard<py::gil_scoped_release>(), "Tomek-link keep-mask");
+    m.def("bin_codes", &bin_codes_dev,
+          py::call_guard<py::gil_scoped_release>(), "Quantile-bin codes");
     m.def("scaler_fit_transform", &scaler_fit_transform_dev,
+          py::call_guard<py::gil_scoped_release>(),
           "StandardScaler fit_transform");
     m.def("pca_fit_transform", &pca_fit_transform_dev,
+          py::call_guard<py::gil_scoped_release>(),
           "Full PCA fit_transform");
 }
