@@ -12,6 +12,7 @@
 #include "forest.hip"
 #include "knn_balance.hip"
 #include "scaler_pca.hip"
+#include "treeshap.hip"
 
 #include <vector>
 
@@ -290,7 +291,47 @@ at::Tensor pca_fit_transform_dev(at::Tensor X, int64_t F) {
     return T;
 }
 
+// ---------------------------------------------------------------------------
+// treeshap
+// ---------------------------------------------------------------------------
+at::Tensor treeshap(at::Tensor codes, at::Tensor j_node_off,
+                    at::Tensor nfeat, at::Tensor nsplit, at::Tensor nleft,
+                    at::Tensor ncnt0, at::Tensor ncnt1) {
+    const at::cuda::OptionalCUDAGuard guard(codes.device());
+    const int n_samples = codes.size(0);
+    const int n_trees = j_node_off.size(0);
+    hipStream_t s = current_stream();
+
+    auto depth = at::zeros({n_trees}, codes.options().dtype(at::kInt));
+    tree_depth_kernel<<<(n_trees + 63) / 64, 64, 0, s>>>(
+        j_node_off.data_ptr<long>(), nfeat.data_ptr<int>(),
+        nleft.data_ptr<int>(), n_trees, depth.data_ptr<int>());
+    const int d_max = depth.max().to(at::kCPU).item<int>() + 1;
+
+    const int GRID = 128;
+    const long n_threads = (long)GRID * SHAP_BLK;
+    const long tri = (long)(d_max + 1) * (d_max + 2) / 2;
+    auto path_ws = at::empty({n_threads * tri * (long)sizeof(PathElem)},
+                             codes.options().dtype(at::kByte));
+    auto frame_ws = at::empty(
+        {n_threads * (d_max + 2) * (long)sizeof(ShapFrame)},
+        codes.options().dtype(at::kByte));
+    auto phi = at::zeros({n_samples, 16},
+                         codes.options().dtype(at::kDouble));
+
+    treeshap_kernel<<<GRID, SHAP_BLK, 0, s>>>(
+        codes.data_ptr<uint8_t>(), n_samples, j_node_off.data_ptr<long>(),
+        nfeat.data_ptr<int>(), nsplit.data_ptr<int>(),
+        nleft.data_ptr<int>(), ncnt0.data_ptr<float>(),
+        ncnt1.data_ptr<float>(), n_trees, d_max,
+        (PathElem*)path_ws.data_ptr(), (ShapFrame*)frame_ws.data_ptr(),
+        phi.data_ptr<double>());
+    return phi;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("treeshap", &treeshap, py::call_guard<py::gil_scoped_release>(),
+          "Path-dependent TreeSHAP (class-0), summed over trees");
     // gil_scoped_release: the forest_fit level loop blocks on stream syncs;
     // releasing the GIL lets other Python threads drive their own streams.
     m.def("forest_fit", &forest_fit,

@@ -303,3 +303,49 @@ class TestCellE2E:
             rt, gt = ref[3], got[3]
             for a, b in zip(rt[:3], gt[:3]):
                 assert abs(a - b) <= max(8, 0.1 * max(a, b)), (keys, rt, gt)
+
+
+class TestTreeShapGpu:
+    def test_treeshap_matches_ref(self, ops, dev):
+        from flake16_framework_amd.models.binning import (
+            bin_codes, compute_bin_cuts,
+        )
+        from flake16_framework_amd.models.forest_ref import (
+            ForestParams, fit_forest,
+        )
+        from flake16_framework_amd.models.treeshap_ref import forest_shap
+
+        X, y = _data(300, f=16, seed=9)
+        cuts = compute_bin_cuts(X)
+        codes = bin_codes(X, cuts)
+        params = ForestParams(10, True, "best", "sqrt", 0)
+        forest = fit_forest(codes, y, params, job_base=40, cuts=cuts)
+        ref = forest_shap(forest, codes[:50], 16)
+
+        codes_d = torch.from_numpy(np.ascontiguousarray(codes)).to(dev)
+        y_d = torch.from_numpy(y).to(dev)
+        j_row_off = torch.zeros(10, dtype=torch.int32, device=dev)
+        j_n = torch.full((10,), len(y), dtype=torch.int32, device=dev)
+        j_key = torch.arange(40, 50, dtype=torch.int32, device=dev)
+        nfeat, nsplit, nleft, ncnt0, ncnt1, j_node_off, _ = ops.forest_fit(
+            codes_d, y_d, j_row_off, j_n, j_key, 16, 4, True, False, 0)
+        phi = ops.treeshap(codes_d[:50].contiguous(), j_node_off, nfeat,
+                           nsplit, nleft, ncnt0, ncnt1)
+        got = phi.cpu().numpy() / 10
+        np.testing.assert_allclose(got[:, :16], ref, atol=1e-9)
+
+    def test_shap_stage_hip_vs_ref(self, ops):
+        from flake16_framework_amd.configgrid import SHAP_CONFIGS
+        from flake16_framework_amd.dataset.synthetic import (
+            make_synthetic_tests,
+        )
+        from flake16_framework_amd.engine.shap_stage import (
+            compute_shap_hip, compute_shap_ref,
+        )
+        tests = make_synthetic_tests(n_tests=250, seed=4)
+        keys = SHAP_CONFIGS[0]
+        ref = compute_shap_ref(keys, 0, tests=tests)
+        got = compute_shap_hip(keys, 0, tests=tests)
+        assert got.shape == ref.shape
+        # scaling runs on device (fp64 reduction order differs) -> tolerance
+        np.testing.assert_allclose(got, ref, atol=5e-4)

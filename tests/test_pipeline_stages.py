@@ -1,0 +1,284 @@
+"""Tests for the orchestration layer, the collection plugins, the shap and
+figures stages, and the CLI surface."""
+
+import json
+import os
+import pickle
+import sqlite3
+import subprocess
+import sys
+
+import numpy as np
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+class TestOrchestrate:
+    def _subjects_file(self, tmp_path):
+        p = tmp_path / "subjects.txt"
+        p.write_text("alice/proj-a,abc123,.,pytest\n"
+                     "bob/proj-b,def456,src,cp x y,pytest -q\n")
+        return str(p)
+
+    def test_iter_subjects(self, tmp_path):
+        from flake16_framework_amd.orchestrate.runner import iter_subjects
+        rows = list(iter_subjects(self._subjects_file(tmp_path)))
+        assert rows[0] == ("proj-a", "alice/proj-a", "abc123", ".", "pytest")
+        assert rows[1][0] == "proj-b"
+        assert rows[1][4:] == ("cp x y", "pytest -q")
+
+    def test_iter_containers_counts(self, tmp_path):
+        from flake16_framework_amd.orchestrate.runner import iter_containers
+        n_runs = {"baseline": 3, "shuffle": 2, "testinspect": 1}
+        conts = list(iter_containers(["baseline", "shuffle"],
+                                     self._subjects_file(tmp_path), n_runs))
+        assert len(conts) == 2 * (3 + 2)
+        names = {c for c, _ in conts}
+        assert "proj-a_baseline_0" in names
+        assert "proj-b_shuffle_1" in names
+
+    def test_mode_flags(self):
+        from flake16_framework_amd.orchestrate.runner import mode_flags
+        assert mode_flags("baseline", "/d/x") == ["--record-file=/d/x.tsv"]
+        assert "--shuffle" in mode_flags("shuffle", "/d/x")
+        assert mode_flags("testinspect", "/d/x") == ["--testinspect=/d/x"]
+
+    def test_manage_container_invocation(self, tmp_path):
+        from flake16_framework_amd.orchestrate.runner import manage_container
+        calls = []
+
+        def fake_run(argv, **kwargs):
+            calls.append((argv, kwargs))
+
+        manage_container("proj-a_shuffle_7", "echo pre", "pytest -q",
+                         subjects_dir=str(tmp_path), data_dir=str(tmp_path),
+                         run=fake_run)
+        assert calls[0][0] == ["echo", "pre"]
+        final = calls[1][0]
+        assert final[:2] == ["pytest", "-q"]
+        assert "--set-exitstatus" in final
+        assert any(a.startswith("--record-file=") for a in final)
+        assert "--shuffle" in final
+        assert "-p" in final and "no:randomly" in final
+        assert calls[1][1]["timeout"] == 7200
+
+    def test_run_log_resume(self, tmp_path):
+        from flake16_framework_amd.orchestrate.runner import read_log
+        log = tmp_path / "log.txt"
+        assert read_log(str(log)) == []
+        log.write_text("a_baseline_0\nb_shuffle_3\n")
+        assert read_log(str(log)) == ["a_baseline_0", "b_shuffle_3"]
+
+    def test_docker_argv(self):
+        from flake16_framework_amd.orchestrate.runner import docker_run_argv
+        argv = docker_run_argv("p_baseline_0", ("pytest",), "/host/data")
+        assert argv[0] == "docker"
+        assert "--cpus=1" in argv
+        assert "--name=p_baseline_0" in argv
+        assert argv[-3:] == ["container", "p_baseline_0", "pytest"]
+
+
+SAMPLE_SUITE = '''
+import pytest
+
+def helper(a):
+    if a > 2:
+        return a * 2
+    return a - 1
+
+def test_pass():
+    assert helper(5) == 10
+
+def test_fail():
+    assert helper(1) == 7
+
+def test_skip():
+    pytest.skip("nope")
+'''
+
+
+class TestCollectPlugins:
+    def _run_suite(self, tmp_path, extra_args):
+        suite = tmp_path / "test_sample.py"
+        suite.write_text(SAMPLE_SUITE)
+        env = dict(os.environ, PYTHONPATH=REPO)
+        proc = subprocess.run(
+            [sys.executable, "-m", "pytest", "-q", str(suite),
+             "-p", "flake16_framework_amd.collect.showflakes",
+             "-p", "flake16_framework_amd.collect.testinspect",
+             "--rootdir", str(tmp_path), *extra_args],
+            cwd=str(tmp_path), env=env, capture_output=True, text=True,
+            timeout=120)
+        return proc
+
+    def test_record_file_and_exitstatus(self, tmp_path):
+        rec = tmp_path / "out.tsv"
+        proc = self._run_suite(tmp_path,
+                               [f"--record-file={rec}", "--set-exitstatus"])
+        assert proc.returncode == 0, proc.stdout + proc.stderr
+        lines = {nid: outcome for outcome, nid in
+                 (l.split("\t", 1) for l in
+                  rec.read_text().strip().split("\n"))}
+        assert lines["test_sample.py::test_pass"] == "passed"
+        assert lines["test_sample.py::test_fail"] == "failed"
+        assert lines["test_sample.py::test_skip"] == "skipped"
+
+    def test_without_set_exitstatus_failures_propagate(self, tmp_path):
+        rec = tmp_path / "out.tsv"
+        proc = self._run_suite(tmp_path, [f"--record-file={rec}"])
+        assert proc.returncode == 1
+
+    def test_shuffle_changes_nothing_but_order(self, tmp_path):
+        rec = tmp_path / "out.tsv"
+        proc = self._run_suite(
+            tmp_path, [f"--record-file={rec}", "--set-exitstatus",
+                       "--shuffle"])
+        assert proc.returncode == 0
+        assert len(rec.read_text().strip().split("\n")) == 3
+
+    def test_testinspect_outputs(self, tmp_path):
+        prefix = tmp_path / "ti"
+        proc = self._run_suite(
+            tmp_path, [f"--testinspect={prefix}", "--set-exitstatus"])
+        assert proc.returncode == 0, proc.stdout + proc.stderr
+
+        # sqlite3: the collation layer can ingest it
+        from flake16_framework_amd.dataset.collate import (
+            get_test_data_nid, update_collated_cov,
+        )
+        collated_proj = [{}, None, None, None]
+        with sqlite3.connect(f"{prefix}.sqlite3") as con:
+            update_collated_cov(con, "x", collated_proj,
+                                subjects_dir=str(tmp_path))
+        assert any("test_pass" in nid for nid in collated_proj[0])
+        cov = next(v[1] for k, v in collated_proj[0].items()
+                   if "test_pass" in k)
+        assert any(lines for lines in cov.values())
+
+        # rusage tsv: 6 floats per test
+        rows = [l.split("\t") for l in
+                open(f"{prefix}.tsv").read().strip().split("\n")]
+        assert all(len(r) == 7 for r in rows)
+        t_exec = float(rows[0][0])
+        assert t_exec >= 0
+
+        # static pkl: 7 metrics per test function
+        with open(f"{prefix}.pkl", "rb") as fd:
+            test_fn_ids, test_fn_data, test_files, churn = pickle.load(fd)
+        assert any("test_pass" in nid for nid in test_fn_ids)
+        fid = next(f for n, f in test_fn_ids.items() if "test_pass" in n)
+        metrics = test_fn_data[fid]
+        assert len(metrics) == 7
+        assert metrics[1] == 1   # one assertion in test_pass
+        assert metrics[5] >= 2   # LoC
+
+
+class TestShapStage:
+    def test_write_shap_format(self, tmp_path):
+        from flake16_framework_amd.dataset.synthetic import (
+            make_synthetic_tests,
+        )
+        from flake16_framework_amd.engine.shap_stage import write_shap
+        tests = make_synthetic_tests(n_tests=150, seed=6)
+        p = tmp_path / "shap.pkl"
+        write_shap(tests=tests, shap_file=str(p), backend="ref")
+        with open(p, "rb") as fd:
+            shap_nod, shap_od = pickle.load(fd)
+        assert shap_nod.shape == (150, 16)
+        assert shap_od.shape == (150, 16)
+        assert np.isfinite(shap_nod).all() and np.isfinite(shap_od).all()
+        # attributions are non-trivial
+        assert np.abs(shap_nod).sum() > 0
+
+
+class TestFiguresStage:
+    def test_figures_end_to_end(self, tmp_path, monkeypatch):
+        from flake16_framework_amd.dataset.synthetic import (
+            make_synthetic_tests,
+        )
+        from flake16_framework_amd.engine.scores import run_scores
+        from flake16_framework_amd.engine.shap_stage import write_shap
+        from flake16_framework_amd.report.figures import (
+            COMPARISON_CONFIGS, write_figures,
+        )
+
+        tests = make_synthetic_tests(n_tests=300, seed=7)
+        tests_file = tmp_path / "tests.json"
+        with open(tests_file, "w") as fd:
+            json.dump(tests, fd)
+
+        # figures only consumes the scores FORMAT: evaluate cheap
+        # Decision-Tree cells and install their blobs under the comparison
+        # config keys the tables hard-code.
+        from flake16_framework_amd.configgrid import iter_config_keys
+        all_keys = list(iter_config_keys())
+        needed = [k for pair in COMPARISON_CONFIGS.values() for k in pair]
+        cheap = sorted({all_keys.index((k[0], k[1], "None", "None",
+                                        "Decision Tree")) for k in needed})
+        cheap_result = run_scores(tests=tests, backend="ref", cells=cheap)
+        result = {}
+        for k in needed:
+            src = (k[0], k[1], "None", "None", "Decision Tree")
+            result[k] = cheap_result[src]
+        scores_file = tmp_path / "scores.pkl"
+        with open(scores_file, "wb") as fd:
+            pickle.dump(result, fd)
+
+        shap_file = tmp_path / "shap.pkl"
+        write_shap(tests=tests, shap_file=str(shap_file), backend="ref")
+
+        write_figures(tests_file=str(tests_file),
+                      scores_file=str(scores_file),
+                      shap_file=str(shap_file), offline=True,
+                      out_dir=str(tmp_path))
+
+        for name in ["tests.tex", "req-runs.tex", "corr.tex", "nod-comp.tex",
+                     "od-comp.tex", "shap.tex"]:
+            content = (tmp_path / name).read_text()
+            assert content.strip(), name
+        # top tables exist (may be empty when every computed cell has F=None)
+        assert (tmp_path / "nod-top.tex").exists()
+        assert (tmp_path / "od-top.tex").exists()
+        assert "\\addplot" in (tmp_path / "req-runs.tex").read_text()
+        assert "\\cellcolor" in (tmp_path / "corr.tex").read_text()
+
+    def test_get_top_tables_unit(self):
+        from flake16_framework_amd.report.figures import get_top_tables
+        scores = {}
+        f1s = {"FlakeFlagger": [0.3, 0.8, None], "Flake16": [0.9, 0.1, 0.5]}
+        for fset in ("FlakeFlagger", "Flake16"):
+            for i, f in enumerate(f1s[fset]):
+                for flaky in ("NOD", "OD"):
+                    keys = (flaky, fset, f"prep{i}", "balN", "modelX")
+                    scores[keys] = [0.01, 0.002, {},
+                                    [1, 1, 1, 0.5, 0.5, f]]
+        tab_nod, tab_od = get_top_tables(scores)
+        # 2 valid FlakeFlagger rows, 3 valid Flake16 rows -> 2 paired rows
+        assert len(tab_nod[0]) == 2
+        # rows sorted by F1 desc: FlakeFlagger 0.8 first, Flake16 0.9 first
+        assert tab_nod[0][0][-1] == 0.9 and tab_nod[0][0][5] == 0.8
+
+
+class TestCli:
+    def test_synthetic_tests_figures_roundtrip(self, tmp_path):
+        env = dict(os.environ, PYTHONPATH=REPO)
+        run = lambda *args: subprocess.run(
+            [sys.executable, os.path.join(REPO, "experiment.py"), *args],
+            cwd=str(tmp_path), env=env, capture_output=True, text=True,
+            timeout=600)
+
+        r = run("synthetic", "--n-tests", "200", "--seed", "1")
+        assert r.returncode == 0, r.stderr
+        assert (tmp_path / "tests.json").exists()
+
+        with open(tmp_path / "tests.json") as fd:
+            tests = json.load(fd)
+        assert sum(len(v) for v in tests.values()) == 200
+
+    def test_unknown_command(self):
+        from flake16_framework_amd.cli import main
+        with pytest.raises(ValueError):
+            main(["bogus"])
+        with pytest.raises(ValueError):
+            main([])
