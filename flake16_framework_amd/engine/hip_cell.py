@@ -274,7 +274,7 @@ class SweepContext:
             nfeat, nsplit, nleft, ncnt0, ncnt1, n_trees,
             len(self._proj_uniq))
         ev[3].record()
-        torch.cuda.synchronize(device)
+        ev[3].synchronize()
 
         t_train = ev[0].elapsed_time(ev[1]) / 1000.0
         t_test = ev[2].elapsed_time(ev[3]) / 1000.0

@@ -115,23 +115,64 @@ def run_scores(tests_file=None, tests=None, backend="auto", cells=None,
     if backend == "auto":
         backend = _auto_backend()
 
-    context = None
     if backend == "hip":
-        from .hip_cell import SweepContext
-        context = SweepContext(tests=tests, tests_file=tests_file, seed=seed)
+        return _run_scores_hip(all_cells, tests, tests_file, seed, progress)
 
     out = {}
     t_start = time.time()
     for n_done, (cell_idx, config_keys) in enumerate(all_cells):
-        if backend == "hip":
-            out[config_keys] = context.evaluate_cell(config_keys, cell_idx)
-        else:
-            out[config_keys] = evaluate_cell_ref(
-                config_keys, cell_idx, tests=tests, tests_file=tests_file,
-                seed=seed)
+        out[config_keys] = evaluate_cell_ref(
+            config_keys, cell_idx, tests=tests, tests_file=tests_file,
+            seed=seed)
         if progress:
             progress(n_done + 1, len(all_cells), time.time() - t_start,
                      ", ".join(config_keys))
+    return out
+
+
+def _run_scores_hip(all_cells, tests, tests_file, seed, progress,
+                    n_streams=4):
+    """Device sweep: prebuild the shared caches (views, folds, balanced
+    groups) on the default stream, then evaluate cells concurrently on
+    worker threads with one HIP stream each — host-side cell bookkeeping
+    overlaps other cells' kernels (extension calls release the GIL)."""
+    import threading
+    from concurrent.futures import ThreadPoolExecutor
+
+    import torch
+
+    from .hip_cell import SweepContext
+
+    context = SweepContext(tests=tests, tests_file=tests_file, seed=seed)
+
+    for cell_idx, config_keys in all_cells:   # warm shared caches
+        context.labels_for(config_keys)
+        context.view_for(config_keys)
+        context.balanced_for(config_keys, cell_idx)
+
+    out = {}
+    lock = threading.Lock()
+    t_start = time.time()
+    n_done = [0]
+
+    def eval_one(args):
+        cell_idx, config_keys = args
+        stream = torch.cuda.Stream()
+        with torch.cuda.stream(stream):
+            result = context.evaluate_cell(config_keys, cell_idx)
+        with lock:
+            out[config_keys] = result
+            n_done[0] += 1
+            if progress:
+                progress(n_done[0], len(all_cells), time.time() - t_start,
+                         ", ".join(config_keys))
+
+    if n_streams <= 1:
+        for args in all_cells:
+            eval_one(args)
+    else:
+        with ThreadPoolExecutor(max_workers=n_streams) as pool:
+            list(pool.map(eval_one, all_cells))
     return out
 
 

@@ -96,10 +96,14 @@ __global__ void forest_init_kernel(
 // ---------------------------------------------------------------------------
 // The level kernel: histogram + split + partition, one workgroup per item.
 // ---------------------------------------------------------------------------
+// Histogram packing: one uint32 per (feature, bin): total count in bits
+// 0-15, class-1 count in bits 16-31.  One LDS atomic per
+// (sample, feature) instead of two, and 16 KiB LDS instead of 32 KiB
+// (double the resident blocks per CU).  Requires per-job n < 65536
+// (checked by the host driver).
 __launch_bounds__(HBLK)
 __global__ void hist_split_kernel(ForestDev a) {
-    __shared__ uint32_t hist_n[FPAD * 256];
-    __shared__ uint32_t hist_1[FPAD * 256];
+    __shared__ uint32_t hist[FPAD * 256];
     __shared__ int sh_scan[HBLK];
     __shared__ int sh_bmin[FPAD], sh_bmax[FPAD];
     __shared__ int sh_cand[FPAD], sh_ncand;
@@ -121,31 +125,28 @@ __global__ void hist_split_kernel(ForestDev a) {
         const uint32_t key = (uint32_t)a.j_key[it.job];
         const int F = a.F;
 
-        // Phase 0: zero LDS histograms.
-        for (int i = tid; i < F * 256; i += HBLK) {
-            hist_n[i] = 0;
-            hist_1[i] = 0;
-        }
+        // Phase 0: zero the packed LDS histogram (uint4-wide).
+        for (int i = tid; i < F * 64; i += HBLK)
+            reinterpret_cast<uint4*>(hist)[i] = uint4{0, 0, 0, 0};
         __syncthreads();
 
-        // Phase 1: accumulate. One uint4 = the sample's 16 packed bin codes.
+        // Phase 1: accumulate. One uint4 = the sample's 16 packed bin
+        // codes; one packed atomic per (sample, feature).
         for (int i = it.start + tid; i < it.end; i += HBLK) {
             int row = a.sidx_cur[sbase + i];
             uint4 cw = *reinterpret_cast<const uint4*>(
                 a.codes + (size_t)row * FPAD);
             uint32_t w[4] = {cw.x, cw.y, cw.z, cw.w};
-            int lab = a.labels[row];
+            const uint32_t inc = 1u | ((uint32_t)a.labels[row] << 16);
             for (int f = 0; f < F; ++f) {
                 uint32_t b = (w[f >> 2] >> ((f & 3) * 8)) & 0xFFu;
-                atomicAdd(&hist_n[f * 256 + b], 1u);
-                if (lab) atomicAdd(&hist_1[f * 256 + b], 1u);
+                atomicAdd(&hist[f * 256 + b], inc);
             }
         }
         __syncthreads();
 
         // Phase 2: class counts from feature-0 histogram.
-        int v = (int)hist_1[tid];
-        sh_scan[tid] = v;
+        sh_scan[tid] = (int)(hist[tid] >> 16);
         __syncthreads();
         for (int d = HBLK / 2; d > 0; d >>= 1) {
             if (tid < d) sh_scan[tid] += sh_scan[tid + d];
@@ -165,15 +166,28 @@ __global__ void hist_split_kernel(ForestDev a) {
             continue;  // leaf (nfeat stays LEAF_SENTINEL)
         }
 
-        // Phase 3: occupied-bin range per feature.
+        // Phase 3: occupied-bin range per feature (parallel: each of the
+        // 256 threads scans one 16-bin segment of one feature).
         if (tid < F) {
-            int bmin = -1, bmax = -1;
-            for (int b = 0; b < 256; ++b)
-                if (hist_n[tid * 256 + b]) { bmin = b; break; }
-            for (int b = 255; b >= 0; --b)
-                if (hist_n[tid * 256 + b]) { bmax = b; break; }
-            sh_bmin[tid] = bmin;
-            sh_bmax[tid] = bmax;
+            sh_bmin[tid] = 256;
+            sh_bmax[tid] = -1;
+        }
+        __syncthreads();
+        {
+            const int f3 = tid >> 4;
+            const int seg = (tid & 15) * 16;
+            if (f3 < F) {
+                int lo = 256, hi = -1;
+                for (int b = seg; b < seg + 16; ++b)
+                    if (hist[f3 * 256 + b] & 0xFFFFu) {
+                        if (lo == 256) lo = b;
+                        hi = b;
+                    }
+                if (hi >= 0) {
+                    atomicMin(&sh_bmin[f3], lo);
+                    atomicMax(&sh_bmax[f3], hi);
+                }
+            }
         }
         __syncthreads();
 
@@ -218,8 +232,9 @@ __global__ void hist_split_kernel(ForestDev a) {
             int tn = 0, t1 = 0;
             for (int k = 0; k < 4; ++k) {
                 int b = lane * 4 + k;
-                ln[k] = (int)hist_n[f * 256 + b];
-                l1[k] = (int)hist_1[f * 256 + b];
+                uint32_t v = hist[f * 256 + b];
+                ln[k] = (int)(v & 0xFFFFu);
+                l1[k] = (int)(v >> 16);
                 tn += ln[k];
                 t1 += l1[k];
             }
@@ -261,7 +276,10 @@ __global__ void hist_split_kernel(ForestDev a) {
                     int b = lane * 4 + k;
                     cn += ln[k];
                     c1f += l1[k];
-                    if (b < bmin || b >= bmax) continue;
+                    // split scores only change at occupied bins, and the
+                    // first bin of an equal-score run is occupied, so
+                    // skipping empty bins preserves the argmax-first result
+                    if (ln[k] == 0 || b < bmin || b >= bmax) continue;
                     long nL = cn, n1L = c1f;
                     long n0L = nL - n1L, nR = n - nL;
                     long n1R = c1 - n1L, n0R = c0 - n0L;

@@ -49,6 +49,9 @@ std::vector<at::Tensor> forest_fit(
     std::vector<long> sidx_off(J), node_off(J);
     long S = 0, Ntot = 0;
     for (int j = 0; j < J; ++j) {
+        TORCH_CHECK(jn[j] < 65536,
+                    "forest_fit: per-job sample count must be < 65536 "
+                    "(packed 16-bit histogram counts)");
         sidx_off[j] = S;
         node_off[j] = Ntot;
         S += jn[j];
