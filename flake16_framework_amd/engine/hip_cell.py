@@ -196,9 +196,88 @@ class SweepContext:
             return _apply_keep(Xs, ys, keep)
         raise ValueError(spec)
 
+    # -- fold-batched balancing (segmented k-NN fills the chip) ------------
+    def _smote_folds(self, Xs, ys, bal_keys):
+        """SMOTE across all folds with one segmented k-NN call; the
+        synthesized rows are bit-identical to per-fold balance.smote."""
+        ops, device, k0 = self.ops, self.device, self.seed
+        metas = []
+        for y in ys:
+            n1 = int(y.sum())
+            n0 = len(y) - n1
+            if n0 == n1:
+                metas.append(None)
+                continue
+            min_label = 1 if n1 < n0 else 0
+            min_rows = np.flatnonzero(y == min_label).astype(np.int32)
+            k = min(5, len(min_rows) - 1)
+            if k < 1:
+                metas.append(None)
+                continue
+            metas.append((min_label, abs(n0 - n1), min_rows, k))
+
+        for kk in sorted({m[3] for m in metas if m}):
+            idxs = [i for i, m in enumerate(metas) if m and m[3] == kk]
+            mins = []
+            seg = [0]
+            for i in idxs:
+                mr = torch.from_numpy(metas[i][2]).to(device)
+                mins.append(Xs[i].index_select(0, mr.long()).contiguous())
+                seg.append(seg[-1] + len(metas[i][2]))
+            Xmin_cat = torch.cat(mins, dim=0).contiguous()
+            seg_off = torch.tensor(seg, dtype=torch.int32, device=device)
+            nn = ops.knn_segmented(Xmin_cat, seg_off, kk, True)
+            for j, i in enumerate(idxs):
+                min_label, n_new, min_rows, k = metas[i]
+                nn_f = nn[seg[j]:seg[j + 1]].contiguous()
+                mr_dev = torch.from_numpy(min_rows).to(device)
+                X_new = ops.smote_interpolate(Xs[i], mr_dev, nn_f, n_new,
+                                              k0, bal_keys[i])
+                Xs[i] = torch.cat([Xs[i], X_new], dim=0).contiguous()
+                ys[i] = np.concatenate(
+                    [ys[i], np.full(n_new, min_label, dtype=np.uint8)])
+        return Xs, ys
+
+    def _clean_folds(self, Xs, ys, kind, clean_all):
+        """ENN / Tomek keep-masks across folds with one segmented k-NN."""
+        ops, device = self.ops, self.device
+        k = 1 if kind == "tomek" else 3
+        active = []
+        for i, y in enumerate(ys):
+            if len(np.unique(y)) < 2 or (kind == "enn" and len(y) <= 3):
+                continue
+            active.append(i)
+        if not active:
+            return Xs, ys
+
+        seg = [0]
+        for i in active:
+            seg.append(seg[-1] + len(ys[i]))
+        X_cat = torch.cat([Xs[i] for i in active], dim=0).contiguous()
+        seg_off = torch.tensor(seg, dtype=torch.int32, device=device)
+        nn = ops.knn_segmented(X_cat, seg_off, k, True)
+
+        for j, i in enumerate(active):
+            y = ys[i]
+            n1 = int(y.sum())
+            maj = 1 if n1 > len(y) - n1 else 0
+            y_dev = torch.from_numpy(y).to(device)
+            nn_f = nn[seg[j]:seg[j + 1]].contiguous()
+            if kind == "tomek":
+                keep = ops.tomek_keep(y_dev, nn_f[:, 0].contiguous(), maj,
+                                      clean_all)
+            else:
+                keep = ops.enn_keep(y_dev, nn_f, 3, maj, clean_all)
+            keep_np = keep.cpu().numpy().astype(bool)
+            idx = torch.from_numpy(
+                np.flatnonzero(keep_np).astype(np.int64)).to(device)
+            Xs[i] = Xs[i].index_select(0, idx).contiguous()
+            ys[i] = y[keep_np]
+        return Xs, ys
+
     def balanced_for(self, keys, cell_idx):
         """Per-fold balanced training codes for the cell's balance group:
-        [(codes_b, yb_len)] x 10 plus fold job keys."""
+        [(codes_b, yb)] x 10.  k-NN work is batched across folds."""
         from .scores import job_ids_for
 
         bk = tuple(keys[:4])
@@ -211,15 +290,30 @@ class SweepContext:
         view = self.view_for(keys)
         X32, F = view["X32"], view["F"]
 
-        out = []
+        Xs, ys, bal_keys = [], [], []
         for i, (train, _) in enumerate(lab["folds"]):
             bal_k1, _ = job_ids_for(keys, cell_idx, i)
+            bal_keys.append(bal_k1)
             tr_idx = torch.from_numpy(train.astype(np.int64)).to(self.device)
-            Xtr = X32.index_select(0, tr_idx).contiguous()
-            Xb, yb = self._balance_dev(Xtr, lab["labels"][train], balancing,
-                                       bal_k1)
-            codes_b = ops.bin_codes(Xb, view["cuts_dev"], view["cut_off"], F)
-            out.append((codes_b, torch.from_numpy(yb).to(self.device)))
+            Xs.append(X32.index_select(0, tr_idx).contiguous())
+            ys.append(lab["labels"][train])
+
+        if balancing in ("smote", "smote+enn", "smote+tomek"):
+            Xs, ys = self._smote_folds(Xs, ys, bal_keys)
+        if balancing == "tomek":
+            Xs, ys = self._clean_folds(Xs, ys, "tomek", False)
+        elif balancing == "enn":
+            Xs, ys = self._clean_folds(Xs, ys, "enn", False)
+        elif balancing == "smote+tomek":
+            Xs, ys = self._clean_folds(Xs, ys, "tomek", True)
+        elif balancing == "smote+enn":
+            Xs, ys = self._clean_folds(Xs, ys, "enn", True)
+
+        out = []
+        for i in range(N_FOLDS):
+            codes_b = ops.bin_codes(Xs[i], view["cuts_dev"], view["cut_off"],
+                                    F)
+            out.append((codes_b, torch.from_numpy(ys[i]).to(self.device)))
         self._balanced[bk] = out
         return out
 

@@ -26,10 +26,11 @@
 
 struct WorkItem {
     int job;
-    int node;    // job-local node id
-    int start;   // job-local sidx range [start, end)
+    int node;       // job-local node id
+    int start;      // job-local sidx range [start, end)
     int end;
     int depth;
+    int hist_slot;  // precomputed-histogram slot in pool[depth&1], or -1
 };
 
 struct ForestDev {
@@ -58,6 +59,15 @@ struct ForestDev {
     int splitter_random;
     uint32_t seed;
     int work_cap;
+    // Histogram-subtraction pools: a splitting node >= hist_save_min
+    // samples accumulates its SMALLER child's histogram itself and derives
+    // the larger child's by subtraction (exact integer counts); both are
+    // stored in pool[(depth+1)&1] and the children skip accumulation.
+    uint32_t* __restrict__ hist_pool0;
+    uint32_t* __restrict__ hist_pool1;
+    int* __restrict__ pool_count;   // [2], write-parity slot zeroed per level
+    int pool_cap;
+    int hist_save_min;
 };
 
 // ---------------------------------------------------------------------------
@@ -89,7 +99,7 @@ __global__ void forest_init_kernel(
     }
     if (threadIdx.x == 0) {
         node_alloc[job] = 1;
-        work[job] = {job, 0, 0, n, 0};
+        work[job] = {job, 0, 0, n, 0, -1};
     }
 }
 
@@ -111,6 +121,9 @@ __global__ void hist_split_kernel(ForestDev a) {
     __shared__ int sh_bin[FPAD], sh_nL[FPAD];
     __shared__ int sh_bestf, sh_bestbin, sh_bestnL;
     __shared__ int sh_loff, sh_roff;
+    __shared__ int sh_lid;                 // allocated left-child node id
+    __shared__ int sh_accum_small;         // phase-8 plan flags
+    __shared__ int sh_slot_small, sh_slot_large, sh_small_is_left;
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -125,22 +138,31 @@ __global__ void hist_split_kernel(ForestDev a) {
         const uint32_t key = (uint32_t)a.j_key[it.job];
         const int F = a.F;
 
-        // Phase 0: zero the packed LDS histogram (uint4-wide).
-        for (int i = tid; i < F * 64; i += HBLK)
-            reinterpret_cast<uint4*>(hist)[i] = uint4{0, 0, 0, 0};
-        __syncthreads();
-
-        // Phase 1: accumulate. One uint4 = the sample's 16 packed bin
-        // codes; one packed atomic per (sample, feature).
-        for (int i = it.start + tid; i < it.end; i += HBLK) {
-            int row = a.sidx_cur[sbase + i];
-            uint4 cw = *reinterpret_cast<const uint4*>(
-                a.codes + (size_t)row * FPAD);
-            uint32_t w[4] = {cw.x, cw.y, cw.z, cw.w};
-            const uint32_t inc = 1u | ((uint32_t)a.labels[row] << 16);
-            for (int f = 0; f < F; ++f) {
-                uint32_t b = (w[f >> 2] >> ((f & 3) * 8)) & 0xFFu;
-                atomicAdd(&hist[f * 256 + b], inc);
+        // Phase 0/1: obtain the node histogram — either load the slot the
+        // parent precomputed (subtraction scheme), or zero + accumulate.
+        if (it.hist_slot >= 0) {
+            const uint32_t* src =
+                ((it.depth & 1) ? a.hist_pool1 : a.hist_pool0)
+                + (size_t)it.hist_slot * (FPAD * 256);
+            for (int i = tid; i < F * 64; i += HBLK)
+                reinterpret_cast<uint4*>(hist)[i] =
+                    reinterpret_cast<const uint4*>(src)[i];
+        } else {
+            for (int i = tid; i < F * 64; i += HBLK)
+                reinterpret_cast<uint4*>(hist)[i] = uint4{0, 0, 0, 0};
+            __syncthreads();
+            // One uint4 = the sample's 16 packed bin codes; one packed
+            // atomic per (sample, feature).
+            for (int i = it.start + tid; i < it.end; i += HBLK) {
+                int row = a.sidx_cur[sbase + i];
+                uint4 cw = *reinterpret_cast<const uint4*>(
+                    a.codes + (size_t)row * FPAD);
+                uint32_t w[4] = {cw.x, cw.y, cw.z, cw.w};
+                const uint32_t inc = 1u | ((uint32_t)a.labels[row] << 16);
+                for (int f = 0; f < F; ++f) {
+                    uint32_t b = (w[f >> 2] >> ((f & 3) * 8)) & 0xFFu;
+                    atomicAdd(&hist[f * 256 + b], inc);
+                }
             }
         }
         __syncthreads();
@@ -327,15 +349,7 @@ __global__ void hist_split_kernel(ForestDev a) {
                 a.nfeat[nbase + it.node] = bf;
                 a.nsplit[nbase + it.node] = bb;
                 a.nleft[nbase + it.node] = l;
-                int qi = atomicAdd(a.nxt_count, 2);
-                if (qi + 1 < a.work_cap) {
-                    a.nxt[qi] = {it.job, l, it.start, it.start + bnl,
-                                 it.depth + 1};
-                    a.nxt[qi + 1] = {it.job, l + 1, it.start + bnl, it.end,
-                                     it.depth + 1};
-                } else {
-                    atomicExch(a.err_flag, 1);
-                }
+                sh_lid = l;
             }
             sh_loff = 0;
             sh_roff = 0;
@@ -385,6 +399,99 @@ __global__ void hist_split_kernel(ForestDev a) {
                 sh_roff += tile_n - tile_left;
             }
             __syncthreads();
+        }
+
+        // Phase 8: histogram-subtraction bookkeeping + child item push.
+        // A big splitting node accumulates its SMALLER child's histogram
+        // here (the children's samples are now contiguous in sidx_nxt) and
+        // stores smaller + (parent - smaller) so both children skip their
+        // accumulation.  Exact integer counts — results are unchanged.
+        if (tid == 0) {
+            // class-1 count of the left child (prefix over chosen feature)
+            int n1L = 0;
+            for (int b = 0; b <= bb; ++b)
+                n1L += (int)(hist[bf * 256 + b] >> 16);
+            const int nR = n - nL;
+            const int n1R = c1 - n1L;
+            const bool l_needs = nL >= 2 && n1L > 0 && n1L < nL;
+            const bool r_needs = nR >= 2 && n1R > 0 && n1R < nR;
+            sh_small_is_left = nL <= nR;
+            const bool small_needs = sh_small_is_left ? l_needs : r_needs;
+            const bool large_needs = sh_small_is_left ? r_needs : l_needs;
+
+            sh_slot_small = sh_slot_large = -1;
+            sh_accum_small = 0;
+            if (n >= a.hist_save_min && (small_needs || large_needs)) {
+                const int wp = (it.depth + 1) & 1;
+                const int want = (small_needs ? 1 : 0) +
+                                 (large_needs ? 1 : 0);
+                int s = atomicAdd(&a.pool_count[wp], want);
+                if (s + want <= a.pool_cap) {
+                    sh_accum_small = 1;
+                    if (small_needs) sh_slot_small = s++;
+                    if (large_needs) sh_slot_large = s;
+                }
+            }
+
+            const int qi = atomicAdd(a.nxt_count, 2);
+            if (qi + 1 < a.work_cap) {
+                const int sl = sh_small_is_left ? sh_slot_small
+                                                : sh_slot_large;
+                const int sr = sh_small_is_left ? sh_slot_large
+                                                : sh_slot_small;
+                a.nxt[qi] = {it.job, sh_lid, it.start, it.start + nL,
+                             it.depth + 1, sh_accum_small ? sl : -1};
+                a.nxt[qi + 1] = {it.job, sh_lid + 1, it.start + nL, it.end,
+                                 it.depth + 1, sh_accum_small ? sr : -1};
+            } else {
+                atomicExch(a.err_flag, 1);
+            }
+        }
+        __syncthreads();
+
+        if (sh_accum_small) {
+            // stash the parent histogram in registers (16 words/thread)
+            uint32_t stash[FPAD];
+            #pragma unroll
+            for (int r = 0; r < FPAD; ++r)
+                stash[r] = (tid + r * HBLK < F * 256)
+                               ? hist[tid + r * HBLK] : 0u;
+            __syncthreads();
+
+            for (int i = tid; i < F * 64; i += HBLK)
+                reinterpret_cast<uint4*>(hist)[i] = uint4{0, 0, 0, 0};
+            __syncthreads();
+
+            const int s0 = sh_small_is_left ? it.start : it.start + nL;
+            const int s1 = sh_small_is_left ? it.start + nL : it.end;
+            for (int i = s0 + tid; i < s1; i += HBLK) {
+                int row = a.sidx_nxt[sbase + i];
+                uint4 cw = *reinterpret_cast<const uint4*>(
+                    a.codes + (size_t)row * FPAD);
+                uint32_t w[4] = {cw.x, cw.y, cw.z, cw.w};
+                const uint32_t inc = 1u | ((uint32_t)a.labels[row] << 16);
+                for (int f = 0; f < F; ++f) {
+                    uint32_t b = (w[f >> 2] >> ((f & 3) * 8)) & 0xFFu;
+                    atomicAdd(&hist[f * 256 + b], inc);
+                }
+            }
+            __syncthreads();
+
+            uint32_t* wpool = ((it.depth + 1) & 1) ? a.hist_pool1
+                                                   : a.hist_pool0;
+            if (sh_slot_small >= 0) {
+                uint32_t* dst = wpool + (size_t)sh_slot_small * (FPAD * 256);
+                for (int i = tid; i < F * 256; i += HBLK)
+                    dst[i] = hist[i];
+            }
+            if (sh_slot_large >= 0) {
+                uint32_t* dst = wpool + (size_t)sh_slot_large * (FPAD * 256);
+                #pragma unroll
+                for (int r = 0; r < FPAD; ++r) {
+                    const int i = tid + r * HBLK;
+                    if (i < F * 256) dst[i] = stash[r] - hist[i];
+                }
+            }
         }
         __syncthreads();
     }

@@ -81,6 +81,83 @@ __global__ void knn_kernel(const float* __restrict__ X,
         for (int j = 0; j < k; ++j) out[(size_t)q * k + j] = bi[j];
 }
 
+// Segmented (fold-batched) k-NN: one call covers many independent
+// candidate sets (e.g. the 10 CV folds of a balance group) so the chip is
+// filled even when each segment alone is small.  seg_off[Nseg+1] bounds the
+// segments in X; seg_blk[Nseg+1] is the prefix of per-segment block counts
+// (ceil(n_seg / KNN_BLK)); outputs are SEGMENT-LOCAL indices — identical
+// bits to a per-segment knn_kernel call.
+__launch_bounds__(KNN_BLK)
+__global__ void knn_segmented_kernel(const float* __restrict__ X,
+                                     const int* __restrict__ seg_off,
+                                     const int* __restrict__ seg_blk,
+                                     int n_seg, int k, int skip_identity,
+                                     int* __restrict__ out) {
+    __shared__ float tile[KNN_BLK][FPAD];
+
+    // locate this block's segment (binary search over seg_blk)
+    int lo = 0, hi = n_seg;
+    while (lo + 1 < hi) {
+        int mid = (lo + hi) >> 1;
+        if (seg_blk[mid] <= (int)blockIdx.x) lo = mid; else hi = mid;
+    }
+    const int seg = lo;
+    const int chunk = blockIdx.x - seg_blk[seg];
+    const int base_row = seg_off[seg];
+    const int n = seg_off[seg + 1] - base_row;
+    const int q = chunk * KNN_BLK + threadIdx.x;   // segment-local query
+
+    float qv[FPAD];
+    if (q < n) {
+        #pragma unroll
+        for (int f = 0; f < FPAD; ++f)
+            qv[f] = X[(size_t)(base_row + q) * FPAD + f];
+    }
+
+    double bd[KMAX];
+    int bi[KMAX];
+    for (int j = 0; j < KMAX; ++j) { bd[j] = 1.0e300; bi[j] = -1; }
+
+    for (int tb = 0; tb < n; tb += KNN_BLK) {
+        const int c = tb + threadIdx.x;
+        if (c < n) {
+            #pragma unroll
+            for (int f = 0; f < FPAD; ++f)
+                tile[threadIdx.x][f] = X[(size_t)(base_row + c) * FPAD + f];
+        }
+        __syncthreads();
+
+        if (q < n) {
+            const int tn = min(KNN_BLK, n - tb);
+            for (int t = 0; t < tn; ++t) {
+                const int cand = tb + t;
+                if (skip_identity && cand == q) continue;
+                double d = 0.0;
+                #pragma unroll
+                for (int f = 0; f < FPAD; ++f) {
+                    double diff = (double)qv[f] - (double)tile[t][f];
+                    d = d + diff * diff;
+                }
+                if (d < bd[k - 1]) {
+                    int j = k - 1;
+                    while (j > 0 && d < bd[j - 1]) {
+                        bd[j] = bd[j - 1];
+                        bi[j] = bi[j - 1];
+                        --j;
+                    }
+                    bd[j] = d;
+                    bi[j] = cand;
+                }
+            }
+        }
+        __syncthreads();
+    }
+
+    if (q < n)
+        for (int j = 0; j < k; ++j)
+            out[(size_t)(base_row + q) * k + j] = bi[j];
+}
+
 // smote_kernel: one thread per synthetic sample.
 // Draw i: pick = bounded(philox(TAG_SMOTE_PICK,0,0,i), n_min*k) ->
 // (row = pick/k, col = pick%k); gap = unit(philox(TAG_SMOTE_GAP,0,0,i)).

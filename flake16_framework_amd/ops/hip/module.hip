@@ -15,6 +15,7 @@
 #include "treeshap.hip"
 
 #include <vector>
+#include <algorithm>
 
 #define CHECK_HIP(expr)                                                     \
     do {                                                                    \
@@ -81,6 +82,17 @@ std::vector<at::Tensor> forest_fit(
     auto counts = at::zeros({2}, opts_i32);
     auto err = at::zeros({1}, opts_i32);
 
+    // Histogram-subtraction pools (see forest.hip): sized for the worst
+    // per-level allocation, 2 slots per splitting node >= HIST_SAVE_MIN.
+    const int HIST_SAVE_MIN = 2048;
+    const long pool_cap =
+        std::min<long>(2 * (S / HIST_SAVE_MIN) + 8, 32768);
+    auto hist_pool0 = at::empty({pool_cap * FPAD * 256},
+                                codes.options().dtype(at::kInt));
+    auto hist_pool1 = at::empty({pool_cap * FPAD * 256},
+                                codes.options().dtype(at::kInt));
+    auto pool_count = at::zeros({2}, opts_i32);
+
     hipStream_t stream = current_stream();
 
     forest_init_kernel<<<J, HBLK, 0, stream>>>(
@@ -116,6 +128,11 @@ std::vector<at::Tensor> forest_fit(
     a.splitter_random = splitter_random ? 1 : 0;
     a.seed = (uint32_t)seed;
     a.work_cap = (int)work_cap;
+    a.hist_pool0 = (uint32_t*)hist_pool0.data_ptr<int>();
+    a.hist_pool1 = (uint32_t*)hist_pool1.data_ptr<int>();
+    a.pool_count = pool_count.data_ptr<int>();
+    a.pool_cap = (int)pool_cap;
+    a.hist_save_min = HIST_SAVE_MIN;
 
     const int GRID = 4096;
     const int CHUNK = 8;
@@ -127,6 +144,9 @@ std::vector<at::Tensor> forest_fit(
             int nx = cur ^ 1;
             CHECK_HIP(hipMemsetAsync(counts.data_ptr<int>() + nx, 0, 4,
                                      stream));
+            // reset the write-parity histogram-pool counter for this level
+            CHECK_HIP(hipMemsetAsync(
+                pool_count.data_ptr<int>() + ((lev + 1) & 1), 0, 4, stream));
             a.sidx_cur = (cur == 0 ? sidx_a : sidx_b).data_ptr<int>();
             a.sidx_nxt = (cur == 0 ? sidx_b : sidx_a).data_ptr<int>();
             a.cur = (const WorkItem*)(cur == 0 ? work_a : work_b).data_ptr();
@@ -191,6 +211,35 @@ at::Tensor knn(at::Tensor X, int64_t k, bool skip_identity) {
     knn_kernel<<<grid, KNN_BLK, 0, current_stream()>>>(
         X.data_ptr<float>(), n, (int)k, skip_identity ? 1 : 0,
         out.data_ptr<int>());
+    return out;
+}
+
+at::Tensor knn_segmented(at::Tensor X, at::Tensor seg_off, int64_t k,
+                         bool skip_identity) {
+    const at::cuda::OptionalCUDAGuard guard(X.device());
+    TORCH_CHECK(X.is_cuda() && X.dtype() == at::kFloat &&
+                X.size(1) == FPAD && X.is_contiguous());
+    TORCH_CHECK(k >= 1 && k <= KMAX);
+    const int n_seg = seg_off.size(0) - 1;
+    const int R = X.size(0);
+
+    auto seg_off_cpu = seg_off.to(at::kCPU);
+    const int* so = seg_off_cpu.data_ptr<int>();
+    std::vector<int> blk(n_seg + 1);
+    blk[0] = 0;
+    for (int s = 0; s < n_seg; ++s) {
+        int n = so[s + 1] - so[s];
+        blk[s + 1] = blk[s] + (n + KNN_BLK - 1) / KNN_BLK;
+    }
+    auto seg_blk = at::from_blob(blk.data(), {n_seg + 1}, at::kInt)
+                       .to(X.device());
+
+    auto out = at::empty({R, k}, X.options().dtype(at::kInt));
+    if (blk[n_seg] > 0)
+        knn_segmented_kernel<<<blk[n_seg], KNN_BLK, 0, current_stream()>>>(
+            X.data_ptr<float>(), seg_off.data_ptr<int>(),
+            seg_blk.data_ptr<int>(), n_seg, (int)k, skip_identity ? 1 : 0,
+            out.data_ptr<int>());
     return out;
 }
 
@@ -345,6 +394,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "Ensemble predict + confusion accumulation");
     m.def("knn", &knn, py::call_guard<py::gil_scoped_release>(),
           "Brute-force k-NN (fp64 distances)");
+    m.def("knn_segmented", &knn_segmented,
+          py::call_guard<py::gil_scoped_release>(),
+          "Fold-batched segmented k-NN (segment-local indices)");
     m.def("smote_interpolate", &smote_interpolate,
           py::call_guard<py::gil_scoped_release>(), "SMOTE synthesis");
     m.def("enn_keep", &enn_keep, py::call_guard<py::gil_scoped_release>(),
