@@ -68,7 +68,15 @@ struct ForestDev {
     int* __restrict__ pool_count;   // [2], write-parity slot zeroed per level
     int pool_cap;
     int hist_save_min;
+    // Small-subtree routing: children with n <= SMALL_N go to this queue
+    // and are finished wholesale by small_subtree_kernel (one wave builds
+    // the whole subtree from register-resident samples).
+    WorkItem* __restrict__ small;
+    int* __restrict__ small_count;
+    int small_cap;
 };
+
+#define SMALL_N 64
 
 // ---------------------------------------------------------------------------
 // Init: fill per-job sample indices (bootstrap or identity) and root items.
@@ -413,8 +421,9 @@ __global__ void hist_split_kernel(ForestDev a) {
                 n1L += (int)(hist[bf * 256 + b] >> 16);
             const int nR = n - nL;
             const int n1R = c1 - n1L;
-            const bool l_needs = nL >= 2 && n1L > 0 && n1L < nL;
-            const bool r_needs = nR >= 2 && n1R > 0 && n1R < nR;
+            // only children that will run the histogram path need a slot
+            const bool l_needs = nL > SMALL_N && n1L > 0 && n1L < nL;
+            const bool r_needs = nR > SMALL_N && n1R > 0 && n1R < nR;
             sh_small_is_left = nL <= nR;
             const bool small_needs = sh_small_is_left ? l_needs : r_needs;
             const bool large_needs = sh_small_is_left ? r_needs : l_needs;
@@ -433,18 +442,34 @@ __global__ void hist_split_kernel(ForestDev a) {
                 }
             }
 
-            const int qi = atomicAdd(a.nxt_count, 2);
-            if (qi + 1 < a.work_cap) {
-                const int sl = sh_small_is_left ? sh_slot_small
-                                                : sh_slot_large;
-                const int sr = sh_small_is_left ? sh_slot_large
-                                                : sh_slot_small;
-                a.nxt[qi] = {it.job, sh_lid, it.start, it.start + nL,
+            const int sl = sh_small_is_left ? sh_slot_small : sh_slot_large;
+            const int sr = sh_small_is_left ? sh_slot_large : sh_slot_small;
+            WorkItem left = {it.job, sh_lid, it.start, it.start + nL,
                              it.depth + 1, sh_accum_small ? sl : -1};
-                a.nxt[qi + 1] = {it.job, sh_lid + 1, it.start + nL, it.end,
-                                 it.depth + 1, sh_accum_small ? sr : -1};
-            } else {
-                atomicExch(a.err_flag, 1);
+            WorkItem right = {it.job, sh_lid + 1, it.start + nL, it.end,
+                              it.depth + 1, sh_accum_small ? sr : -1};
+            // children small enough for the wave-subtree path skip the
+            // level queue entirely
+            WorkItem* q[2];
+            int qn = 0, sn = 0;
+            WorkItem* sq[2];
+            if (nL <= SMALL_N) sq[sn++] = &left; else q[qn++] = &left;
+            if (nR <= SMALL_N) sq[sn++] = &right; else q[qn++] = &right;
+            if (qn) {
+                const int qi = atomicAdd(a.nxt_count, qn);
+                if (qi + qn <= a.work_cap) {
+                    for (int x = 0; x < qn; ++x) a.nxt[qi + x] = *q[x];
+                } else {
+                    atomicExch(a.err_flag, 1);
+                }
+            }
+            if (sn) {
+                const int si = atomicAdd(a.small_count, sn);
+                if (si + sn <= a.small_cap) {
+                    for (int x = 0; x < sn; ++x) a.small[si + x] = *sq[x];
+                } else {
+                    atomicExch(a.err_flag, 1);
+                }
             }
         }
         __syncthreads();
@@ -494,6 +519,194 @@ __global__ void hist_split_kernel(ForestDev a) {
             }
         }
         __syncthreads();
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Wave-per-subtree builder for nodes with n <= SMALL_N (=64) samples.
+//
+// One 64-lane wave finishes the WHOLE subtree: each lane holds one
+// sample's 16 packed codes + label in registers; node membership is a
+// 64-bit lane mask; class/prefix counts come from __ballot+popcount; the
+// DFS stack lives in LDS.  No histograms, no sample-index traffic, no
+// further level round-trips.  All randomness stays keyed on the node's
+// (start, end) range (child ranges follow the stable-partition arithmetic
+// [s, s+nL) / [s+nL, e)), and split scores use the identical fp64
+// expression — trees are bit-identical to the histogram path / numpy
+// reference.
+// ---------------------------------------------------------------------------
+struct SmallFrame {
+    unsigned long long mask;
+    int s, e, depth, node;
+};
+
+#define SMALL_STACK 68
+
+__launch_bounds__(HBLK)
+__global__ void small_subtree_kernel(ForestDev a,
+                                     const int* __restrict__ sidx_level) {
+    __shared__ SmallFrame stack_ws[HBLK / 64][SMALL_STACK];
+    __shared__ int perm_ws[HBLK / 64][FPAD];
+
+    const int lane = threadIdx.x & 63;
+    const int wave = threadIdx.x >> 6;
+    const int waves_total = gridDim.x * (HBLK / 64);
+    const int wave_id = blockIdx.x * (HBLK / 64) + wave;
+    const int n_items = *a.small_count;
+    SmallFrame* stack = stack_ws[wave];
+    int* perm = perm_ws[wave];
+
+    for (int wi = wave_id; wi < n_items; wi += waves_total) {
+        const WorkItem it = a.small[min(wi, a.small_cap - 1)];
+        const int n0 = it.end - it.start;
+        const long sbase = a.j_sidx_off[it.job];
+        const long nbase = a.j_node_off[it.job];
+        const uint32_t key = (uint32_t)a.j_key[it.job];
+        const int F = a.F;
+
+        // lane -> sample (register-resident)
+        uint32_t w[4] = {0, 0, 0, 0};
+        int label = 0;
+        if (lane < n0) {
+            const int row = sidx_level[sbase + it.start + lane];
+            uint4 cw = *reinterpret_cast<const uint4*>(
+                a.codes + (size_t)row * FPAD);
+            w[0] = cw.x; w[1] = cw.y; w[2] = cw.z; w[3] = cw.w;
+            label = a.labels[row];
+        }
+        const unsigned long long lab_mask = __ballot(label != 0);
+
+        int sp = 0;
+        if (lane == 0)
+            stack[0] = {n0 >= 64 ? ~0ULL : ((1ULL << n0) - 1ULL),
+                        it.start, it.end, it.depth, it.node};
+
+        while (sp >= 0) {
+            // pop (lane-uniform: all lanes read the same LDS entry)
+            const unsigned long long mask = stack[sp].mask;
+            const int s = stack[sp].s;
+            const int e = stack[sp].e;
+            const int depth = stack[sp].depth;
+            const int node = stack[sp].node;
+            --sp;
+
+            const int n = __popcll(mask);
+            const int c1 = __popcll(mask & lab_mask);
+            const int c0 = n - c1;
+            if (lane == 0) {
+                a.ncnt0[nbase + node] = (float)c0;
+                a.ncnt1[nbase + node] = (float)c1;
+            }
+            if (n < 2 || c0 == 0 || c1 == 0) continue;
+
+            // feature permutation (lane 0, broadcast via LDS)
+            if (lane == 0) {
+                for (int f = 0; f < F; ++f) perm[f] = f;
+                uint32_t tag = TAG_FEATSEL |
+                               ((uint32_t)(depth & 0xFF) << 8);
+                for (int i = 0; i < F - 1; ++i) {
+                    uint32_t u = philox_draw(tag, (uint32_t)s, (uint32_t)e,
+                                             (uint32_t)i, a.seed, key);
+                    int j = i + (int)philox_bounded(u, (uint32_t)(F - i));
+                    int t = perm[i]; perm[i] = perm[j]; perm[j] = t;
+                }
+            }
+            // lane 0's LDS writes are visible to the wave in program order
+
+            double best_s = -1.0e300;
+            int best_f = -1, best_b = -1, best_nl = 0;
+            int n_eval = 0;
+
+            for (int pi = 0; pi < F && n_eval < a.max_features; ++pi) {
+                const int f = perm[pi];
+                const int my_code = (int)((w[f >> 2] >> ((f & 3) * 8))
+                                          & 0xFFu);
+                const bool in = (mask >> lane) & 1ULL;
+                // occupied-bin range via wave min/max (inactive lanes
+                // neutralized)
+                int cmin = in ? my_code : 256;
+                int cmax = in ? my_code : -1;
+                for (int d = 32; d > 0; d >>= 1) {
+                    cmin = min(cmin, __shfl_xor(cmin, d));
+                    cmax = max(cmax, __shfl_xor(cmax, d));
+                }
+                if (cmin == cmax) continue;   // constant: not counted
+                ++n_eval;
+
+                if (a.splitter_random) {
+                    uint32_t tag = TAG_THRESH |
+                                   ((uint32_t)(depth & 0xFF) << 8);
+                    uint32_t u = philox_draw(tag, (uint32_t)s, (uint32_t)e,
+                                             (uint32_t)f, a.seed, key);
+                    const int b = cmin + (int)philox_bounded(
+                        u, (uint32_t)(cmax - cmin));
+                    const unsigned long long lm =
+                        mask & __ballot(in && my_code <= b);
+                    const long nL = __popcll(lm);
+                    const long n1L = __popcll(lm & lab_mask);
+                    const long n0L = nL - n1L, nR = n - nL;
+                    const long n1R = c1 - n1L, n0R = c0 - n0L;
+                    if (nL > 0 && nR > 0) {
+                        double sc = (double)(n0L * n0L + n1L * n1L)
+                                        / (double)nL
+                                    + (double)(n0R * n0R + n1R * n1R)
+                                        / (double)nR;
+                        if (sc > best_s) {
+                            best_s = sc; best_f = f; best_b = b;
+                            best_nl = (int)nL;
+                        }
+                    }
+                } else {
+                    // iterate occupied code values ascending
+                    int v = cmin;
+                    while (v < cmax) {
+                        const unsigned long long lm =
+                            mask & __ballot(in && my_code <= v);
+                        const long nL = __popcll(lm);
+                        const long n1L = __popcll(lm & lab_mask);
+                        const long n0L = nL - n1L, nR = n - nL;
+                        const long n1R = c1 - n1L, n0R = c0 - n0L;
+                        double sc = (double)(n0L * n0L + n1L * n1L)
+                                        / (double)nL
+                                    + (double)(n0R * n0R + n1R * n1R)
+                                        / (double)nR;
+                        if (sc > best_s) {
+                            best_s = sc; best_f = f; best_b = v;
+                            best_nl = (int)nL;
+                        }
+                        // next occupied value > v
+                        int nxt = (in && my_code > v) ? my_code : 256;
+                        for (int d = 32; d > 0; d >>= 1)
+                            nxt = min(nxt, __shfl_xor(nxt, d));
+                        v = nxt;
+                    }
+                }
+            }
+
+            if (best_f < 0) continue;   // no valid split: leaf
+
+            int lid = 0;
+            if (lane == 0) {
+                lid = atomicAdd(&a.node_alloc[it.job], 2);
+                a.nfeat[nbase + node] = best_f;
+                a.nsplit[nbase + node] = best_b;
+                a.nleft[nbase + node] = lid;
+            }
+            lid = __shfl(lid, 0);
+
+            const int bc = (int)((w[best_f >> 2] >> ((best_f & 3) * 8))
+                                 & 0xFFu);
+            const unsigned long long lmask =
+                mask & __ballot(((mask >> lane) & 1ULL) && bc <= best_b);
+            const int nL = best_nl;
+
+            if (lane == 0) {
+                stack[sp + 1] = {mask & ~lmask, s + nL, e, depth + 1,
+                                 lid + 1};
+                stack[sp + 2] = {lmask, s, s + nL, depth + 1, lid};
+            }
+            sp += 2;   // left child on top: DFS order (order immaterial)
+        }
     }
 }
 

@@ -82,6 +82,10 @@ std::vector<at::Tensor> forest_fit(
     auto counts = at::zeros({2}, opts_i32);
     auto err = at::zeros({1}, opts_i32);
 
+    auto small_q = at::empty({(S + 2) * (long)sizeof(WorkItem)},
+                             codes.options().dtype(at::kByte));
+    auto small_count = at::zeros({1}, opts_i32);
+
     // Histogram-subtraction pools (see forest.hip): sized for the worst
     // per-level allocation, 2 slots per splitting node >= HIST_SAVE_MIN.
     const int HIST_SAVE_MIN = 2048;
@@ -133,6 +137,9 @@ std::vector<at::Tensor> forest_fit(
     a.pool_count = pool_count.data_ptr<int>();
     a.pool_cap = (int)pool_cap;
     a.hist_save_min = HIST_SAVE_MIN;
+    a.small = (WorkItem*)small_q.data_ptr();
+    a.small_count = small_count.data_ptr<int>();
+    a.small_cap = (int)(S + 2);
 
     const int GRID = 4096;
     const int CHUNK = 8;
@@ -153,7 +160,11 @@ std::vector<at::Tensor> forest_fit(
             a.nxt = (WorkItem*)(cur == 0 ? work_b : work_a).data_ptr();
             a.cur_count = counts.data_ptr<int>() + cur;
             a.nxt_count = counts.data_ptr<int>() + nx;
+            CHECK_HIP(hipMemsetAsync(small_count.data_ptr<int>(), 0, 4,
+                                     stream));
             hist_split_kernel<<<GRID, HBLK, 0, stream>>>(a);
+            small_subtree_kernel<<<2048, HBLK, 0, stream>>>(
+                a, a.sidx_nxt);
             CHECK_HIP(hipMemcpyAsync(pinned_p + (c % PINSZ),
                                      counts.data_ptr<int>() + nx, 4,
                                      hipMemcpyDeviceToHost, stream));
