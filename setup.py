@@ -23,6 +23,11 @@ setup(
         CUDAExtension(
             name="flake16_framework_amd.ops._hip",
             sources=["flake16_framework_amd/ops/hip/module.hip"],
+            depends=["flake16_framework_amd/ops/hip/forest.hip",
+                     "flake16_framework_amd/ops/hip/knn_balance.hip",
+                     "flake16_framework_amd/ops/hip/scaler_pca.hip",
+                     "flake16_framework_amd/ops/hip/treeshap.hip",
+                     "flake16_framework_amd/ops/hip/philox.h"],
             extra_compile_args={
                 "cxx": ["-O3"],
                 "nvcc": ["-O3", "-ffp-contract=off"],
