@@ -46,7 +46,9 @@ def main():
     rank, world = comm.init_from_env()
     use_cuda = torch.cuda.is_available() and args.backend == "hip"
     if use_cuda:
-        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
+        # clamp so oversubscribed smoke runs (2 ranks, 1 GPU) still work
+        torch.cuda.set_device(min(int(os.environ.get("LOCAL_RANK", "0")),
+                                  torch.cuda.device_count() - 1))
 
     tests = make_synthetic_tests(n_tests=args.n_tests, seed=0)
     my_cells = comm.shard_cells(world, rank, n_cells=args.cells)

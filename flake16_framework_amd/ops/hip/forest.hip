@@ -175,14 +175,15 @@ __global__ void hist_split_kernel(ForestDev a) {
         }
         __syncthreads();
 
-        // Phase 2: class counts from feature-0 histogram.
-        sh_scan[tid] = (int)(hist[tid] >> 16);
-        __syncthreads();
-        for (int d = HBLK / 2; d > 0; d >>= 1) {
-            if (tid < d) sh_scan[tid] += sh_scan[tid + d];
-            __syncthreads();
+        // Phase 2: class counts from feature-0 histogram (wave reduce +
+        // one cross-wave combine: 1 barrier instead of 8).
+        {
+            int v = (int)(hist[tid] >> 16);
+            for (int d = 32; d > 0; d >>= 1) v += __shfl_down(v, d);
+            if (lane == 0) sh_scan[wave] = v;
         }
-        const int c1 = sh_scan[0];
+        __syncthreads();
+        const int c1 = sh_scan[0] + sh_scan[1] + sh_scan[2] + sh_scan[3];
         const int c0 = n - c1;
         __syncthreads();
 
@@ -373,6 +374,9 @@ __global__ void hist_split_kernel(ForestDev a) {
         const int nL = sh_bestnL;
 
         // Phase 7: stable partition into sidx_nxt, 256-wide tiles.
+        // Ranks come from wave ballots (valid lanes are a prefix of the
+        // tile, so cross-wave offsets are 4 LDS words): 2 barriers per
+        // tile instead of 16.
         for (int base = it.start; base < it.end; base += HBLK) {
             const int i = base + tid;
             const bool valid = i < it.end;
@@ -382,24 +386,29 @@ __global__ void hist_split_kernel(ForestDev a) {
                 uint32_t b = a.codes[(size_t)row * FPAD + bf];
                 flag = (int)(b <= (uint32_t)bb);
             }
-            sh_scan[tid] = flag;
+            const unsigned long long lm = __ballot(valid && flag);
+            const unsigned long long below = (1ULL << lane) - 1ULL;
+            const int left_rank = __popcll(lm & below);
+            const int wave_left = __popcll(lm);
+            if (lane == 0) sh_scan[wave] = wave_left;
             __syncthreads();
-            for (int d = 1; d < HBLK; d <<= 1) {
-                int t = (tid >= d) ? sh_scan[tid - d] : 0;
-                __syncthreads();
-                sh_scan[tid] += t;
-                __syncthreads();
-            }
-            const int incl = sh_scan[tid];
-            const int tile_left = sh_scan[HBLK - 1];
+            int wave_left_excl = 0;
+            for (int ww = 0; ww < wave; ++ww)
+                wave_left_excl += sh_scan[ww];
+            const int tile_left = sh_scan[0] + sh_scan[1] + sh_scan[2]
+                                  + sh_scan[3];
             const int tile_n = min(HBLK, it.end - base);
             if (valid) {
-                const int excl = incl - flag;
-                if (flag)
-                    a.sidx_nxt[sbase + it.start + sh_loff + excl] = row;
-                else
+                if (flag) {
+                    a.sidx_nxt[sbase + it.start + sh_loff + wave_left_excl
+                               + left_rank] = row;
+                } else {
+                    // rights before me = my tile index - lefts before me
+                    const int rights_before =
+                        (i - base) - (wave_left_excl + left_rank);
                     a.sidx_nxt[sbase + it.start + nL + sh_roff
-                               + (i - base) - excl] = row;
+                               + rights_before] = row;
+                }
             }
             __syncthreads();
             if (tid == 0) {
@@ -516,6 +525,267 @@ __global__ void hist_split_kernel(ForestDev a) {
                     const int i = tid + r * HBLK;
                     if (i < F * 256) dst[i] = stash[r] - hist[i];
                 }
+            }
+        }
+        __syncthreads();
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Extra-Trees level kernel: histogram-free.
+//
+// The random splitter needs only (a) per-feature occupied min/max codes to
+// know constancy and the draw range, and (b) left counts at ONE drawn bin
+// per candidate feature — two atomic-free passes over the node's samples
+// instead of a 16-way-atomic histogram pass.  Scores, draws and
+// tie-breaking are identical to the histogram path (same Philox counters,
+// same fp64 expression), so trees are unchanged.  Used for all
+// splitter_random jobs; the wave-subtree kernel handles their <=64 tails.
+// ---------------------------------------------------------------------------
+__launch_bounds__(HBLK)
+__global__ void et_split_kernel(ForestDev a) {
+    __shared__ int sh_scan[HBLK];
+    __shared__ int sh_min[FPAD], sh_max[FPAD];
+    __shared__ int sh_cand[FPAD], sh_cbin[FPAD], sh_ncand;
+    __shared__ int sh_cnt[4][FPAD], sh_cnt1[4][FPAD];
+    __shared__ int sh_bestf, sh_bestbin, sh_bestnL;
+    __shared__ int sh_loff, sh_roff, sh_lid;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int n_items = *a.cur_count;
+
+    for (int wi = blockIdx.x; wi < n_items; wi += gridDim.x) {
+        WorkItem it = a.cur[wi];
+        const int n = it.end - it.start;
+        const long sbase = a.j_sidx_off[it.job];
+        const long nbase = a.j_node_off[it.job];
+        const uint32_t key = (uint32_t)a.j_key[it.job];
+        const int F = a.F;
+
+        if (tid < F) {
+            sh_min[tid] = 256;
+            sh_max[tid] = -1;
+        }
+        __syncthreads();
+
+        // Pass 1: per-feature min/max + class-1 count.
+        int lmin[FPAD], lmax[FPAD];
+        #pragma unroll
+        for (int f = 0; f < FPAD; ++f) { lmin[f] = 256; lmax[f] = -1; }
+        int lc1 = 0;
+        for (int i = it.start + tid; i < it.end; i += HBLK) {
+            int row = a.sidx_cur[sbase + i];
+            uint4 cw = *reinterpret_cast<const uint4*>(
+                a.codes + (size_t)row * FPAD);
+            uint32_t w[4] = {cw.x, cw.y, cw.z, cw.w};
+            lc1 += a.labels[row];
+            for (int f = 0; f < F; ++f) {
+                const int b = (int)((w[f >> 2] >> ((f & 3) * 8)) & 0xFFu);
+                lmin[f] = min(lmin[f], b);
+                lmax[f] = max(lmax[f], b);
+            }
+        }
+        for (int d = 32; d > 0; d >>= 1) lc1 += __shfl_down(lc1, d);
+        if (lane == 0) sh_scan[wave] = lc1;
+        for (int f = 0; f < F; ++f) {
+            int mn = lmin[f], mx = lmax[f];
+            for (int d = 32; d > 0; d >>= 1) {
+                mn = min(mn, __shfl_xor(mn, d));
+                mx = max(mx, __shfl_xor(mx, d));
+            }
+            if (lane == 0) {
+                atomicMin(&sh_min[f], mn);
+                atomicMax(&sh_max[f], mx);
+            }
+        }
+        __syncthreads();
+        const int c1 = sh_scan[0] + sh_scan[1] + sh_scan[2] + sh_scan[3];
+        const int c0 = n - c1;
+        __syncthreads();
+
+        if (tid == 0) {
+            a.ncnt0[nbase + it.node] = (float)c0;
+            a.ncnt1[nbase + it.node] = (float)c1;
+        }
+        if (n < 2 || c0 == 0 || c1 == 0) {
+            __syncthreads();
+            continue;
+        }
+
+        // Permutation walk + threshold draws (thread 0).
+        if (tid == 0) {
+            int perm[FPAD];
+            for (int f = 0; f < F; ++f) perm[f] = f;
+            uint32_t tag = TAG_FEATSEL | ((uint32_t)(it.depth & 0xFF) << 8);
+            for (int i = 0; i < F - 1; ++i) {
+                uint32_t u = philox_draw(tag, (uint32_t)it.start,
+                                         (uint32_t)it.end, (uint32_t)i,
+                                         a.seed, key);
+                int j = i + (int)philox_bounded(u, (uint32_t)(F - i));
+                int t = perm[i]; perm[i] = perm[j]; perm[j] = t;
+            }
+            uint32_t ttag = TAG_THRESH | ((uint32_t)(it.depth & 0xFF) << 8);
+            int nc = 0;
+            for (int i = 0; i < F && nc < a.max_features; ++i) {
+                int f = perm[i];
+                if (sh_min[f] == sh_max[f]) continue;
+                uint32_t u = philox_draw(ttag, (uint32_t)it.start,
+                                         (uint32_t)it.end, (uint32_t)f,
+                                         a.seed, key);
+                sh_cand[nc] = f;
+                sh_cbin[nc] = sh_min[f] + (int)philox_bounded(
+                    u, (uint32_t)(sh_max[f] - sh_min[f]));
+                ++nc;
+            }
+            sh_ncand = nc;
+        }
+        __syncthreads();
+
+        const int ncand = sh_ncand;
+        if (ncand == 0) {
+            __syncthreads();
+            continue;
+        }
+
+        // Pass 2: left counts at each candidate's drawn bin.
+        int cnt[FPAD], cnt1[FPAD];
+        for (int ci = 0; ci < ncand; ++ci) { cnt[ci] = 0; cnt1[ci] = 0; }
+        for (int i = it.start + tid; i < it.end; i += HBLK) {
+            int row = a.sidx_cur[sbase + i];
+            uint4 cw = *reinterpret_cast<const uint4*>(
+                a.codes + (size_t)row * FPAD);
+            uint32_t w[4] = {cw.x, cw.y, cw.z, cw.w};
+            const int lab = a.labels[row];
+            for (int ci = 0; ci < ncand; ++ci) {
+                const int f = sh_cand[ci];
+                const int b = (int)((w[f >> 2] >> ((f & 3) * 8)) & 0xFFu);
+                const int le = b <= sh_cbin[ci];
+                cnt[ci] += le;
+                cnt1[ci] += le & lab;
+            }
+        }
+        for (int ci = 0; ci < ncand; ++ci) {
+            int c = cnt[ci], c1l = cnt1[ci];
+            for (int d = 32; d > 0; d >>= 1) {
+                c += __shfl_down(c, d);
+                c1l += __shfl_down(c1l, d);
+            }
+            if (lane == 0) {
+                sh_cnt[wave][ci] = c;
+                sh_cnt1[wave][ci] = c1l;
+            }
+        }
+        __syncthreads();
+
+        // Select (thread 0): candidates in perm order, strict >.
+        if (tid == 0) {
+            double best_s = -1.0e300;
+            int bf = -1, bb = -1, bnl = 0;
+            for (int ci = 0; ci < ncand; ++ci) {
+                const long nL = sh_cnt[0][ci] + sh_cnt[1][ci]
+                                + sh_cnt[2][ci] + sh_cnt[3][ci];
+                const long n1L = sh_cnt1[0][ci] + sh_cnt1[1][ci]
+                                 + sh_cnt1[2][ci] + sh_cnt1[3][ci];
+                const long n0L = nL - n1L, nR = n - nL;
+                const long n1R = c1 - n1L, n0R = c0 - n0L;
+                if (nL == 0 || nR == 0) continue;
+                double sc = (double)(n0L * n0L + n1L * n1L) / (double)nL
+                            + (double)(n0R * n0R + n1R * n1R) / (double)nR;
+                if (sc > best_s) {
+                    best_s = sc;
+                    bf = sh_cand[ci];
+                    bb = sh_cbin[ci];
+                    bnl = (int)nL;
+                }
+            }
+            sh_bestf = bf;
+            sh_bestbin = bb;
+            sh_bestnL = bnl;
+            if (bf >= 0) {
+                int l = atomicAdd(&a.node_alloc[it.job], 2);
+                a.nfeat[nbase + it.node] = bf;
+                a.nsplit[nbase + it.node] = bb;
+                a.nleft[nbase + it.node] = l;
+                sh_lid = l;
+            }
+            sh_loff = 0;
+            sh_roff = 0;
+        }
+        __syncthreads();
+
+        const int bf = sh_bestf;
+        if (bf < 0) {
+            __syncthreads();
+            continue;
+        }
+        const int bb = sh_bestbin;
+        const int nL = sh_bestnL;
+
+        // Partition (same scheme as hist_split_kernel phase 7).
+        for (int base = it.start; base < it.end; base += HBLK) {
+            const int i = base + tid;
+            const bool valid = i < it.end;
+            int row = 0, flag = 0;
+            if (valid) {
+                row = a.sidx_cur[sbase + i];
+                uint32_t b = a.codes[(size_t)row * FPAD + bf];
+                flag = (int)(b <= (uint32_t)bb);
+            }
+            const unsigned long long lm = __ballot(valid && flag);
+            const unsigned long long below = (1ULL << lane) - 1ULL;
+            const int left_rank = __popcll(lm & below);
+            if (lane == 0) sh_scan[wave] = __popcll(lm);
+            __syncthreads();
+            int wave_left_excl = 0;
+            for (int ww = 0; ww < wave; ++ww)
+                wave_left_excl += sh_scan[ww];
+            const int tile_left = sh_scan[0] + sh_scan[1] + sh_scan[2]
+                                  + sh_scan[3];
+            const int tile_n = min(HBLK, it.end - base);
+            if (valid) {
+                if (flag)
+                    a.sidx_nxt[sbase + it.start + sh_loff + wave_left_excl
+                               + left_rank] = row;
+                else
+                    a.sidx_nxt[sbase + it.start + nL + sh_roff
+                               + (i - base) - (wave_left_excl + left_rank)]
+                        = row;
+            }
+            __syncthreads();
+            if (tid == 0) {
+                sh_loff += tile_left;
+                sh_roff += tile_n - tile_left;
+            }
+            __syncthreads();
+        }
+
+        // Push children (no histogram slots on the ET path).
+        if (tid == 0) {
+            WorkItem left = {it.job, sh_lid, it.start, it.start + nL,
+                             it.depth + 1, -1};
+            WorkItem right = {it.job, sh_lid + 1, it.start + nL, it.end,
+                              it.depth + 1, -1};
+            WorkItem* q[2];
+            int qn = 0, sn = 0;
+            WorkItem* sq[2];
+            if (nL <= SMALL_N) sq[sn++] = &left; else q[qn++] = &left;
+            if (it.end - it.start - nL <= SMALL_N) sq[sn++] = &right;
+            else q[qn++] = &right;
+            if (qn) {
+                const int qi = atomicAdd(a.nxt_count, qn);
+                if (qi + qn <= a.work_cap)
+                    for (int x = 0; x < qn; ++x) a.nxt[qi + x] = *q[x];
+                else
+                    atomicExch(a.err_flag, 1);
+            }
+            if (sn) {
+                const int si = atomicAdd(a.small_count, sn);
+                if (si + sn <= a.small_cap)
+                    for (int x = 0; x < sn; ++x) a.small[si + x] = *sq[x];
+                else
+                    atomicExch(a.err_flag, 1);
             }
         }
         __syncthreads();

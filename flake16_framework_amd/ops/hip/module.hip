@@ -162,7 +162,10 @@ std::vector<at::Tensor> forest_fit(
             a.nxt_count = counts.data_ptr<int>() + nx;
             CHECK_HIP(hipMemsetAsync(small_count.data_ptr<int>(), 0, 4,
                                      stream));
-            hist_split_kernel<<<GRID, HBLK, 0, stream>>>(a);
+            if (splitter_random)
+                et_split_kernel<<<GRID, HBLK, 0, stream>>>(a);
+            else
+                hist_split_kernel<<<GRID, HBLK, 0, stream>>>(a);
             small_subtree_kernel<<<2048, HBLK, 0, stream>>>(
                 a, a.sidx_nxt);
             CHECK_HIP(hipMemcpyAsync(pinned_p + (c % PINSZ),
@@ -371,7 +374,7 @@ at::Tensor treeshap(at::Tensor codes, at::Tensor j_node_off,
         nleft.data_ptr<int>(), n_trees, depth.data_ptr<int>());
     const int d_max = depth.max().to(at::kCPU).item<int>() + 1;
 
-    const int GRID = 128;
+    const int GRID = 512;
     const long n_threads = (long)GRID * SHAP_BLK;
     const long tri = (long)(d_max + 1) * (d_max + 2) / 2;
     auto path_ws = at::empty({n_threads * tri * (long)sizeof(PathElem)},

@@ -45,7 +45,8 @@ def init_from_env(backend=None):
     if backend is None:
         backend = "nccl" if torch.cuda.is_available() else "gloo"
     if backend == "nccl":
-        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
+        torch.cuda.set_device(min(int(os.environ.get("LOCAL_RANK", "0")),
+                                  torch.cuda.device_count() - 1))
     dist().init_process_group(backend=backend)
     return rank_world()
 
