@@ -53,10 +53,15 @@ def main(argv=None):
         write_tests()
     elif command == "scores":
         backend = _pop_flag(args, "--backend", "auto")
+        checkpoint = _pop_flag(args, "--checkpoint", None)
+        trace = _pop_flag(args, "--trace", None)
+        if trace:
+            from .utils.trace import set_trace_file
+            set_trace_file(trace)
         from .engine.scores import write_scores
         from .parallel import comm
         comm.init_from_env()
-        write_scores(backend=backend)
+        write_scores(backend=backend, checkpoint=checkpoint)
     elif command == "shap":
         backend = _pop_flag(args, "--backend", "auto")
         from .engine.shap_stage import write_shap

@@ -103,10 +103,14 @@ def evaluate_cell_ref(config_keys, cell_idx, tests=None, tests_file=None,
 
 
 def run_scores(tests_file=None, tests=None, backend="auto", cells=None,
-               progress=None, seed=GLOBAL_SEED):
+               progress=None, seed=GLOBAL_SEED, on_result=None):
     """Evaluate `cells` (an iterable of (cell_idx, config_keys); default all
     216) and return {config_keys: [t_train, t_test, scores, scores_total]}.
+    on_result(config_keys, value): optional per-cell callback (used by the
+    checkpoint writer).
     """
+    from ..utils.trace import trace_span
+
     all_cells = list(enumerate(iter_config_keys()))
     if cells is not None:
         wanted = set(cells)
@@ -116,14 +120,18 @@ def run_scores(tests_file=None, tests=None, backend="auto", cells=None,
         backend = _auto_backend()
 
     if backend == "hip":
-        return _run_scores_hip(all_cells, tests, tests_file, seed, progress)
+        return _run_scores_hip(all_cells, tests, tests_file, seed, progress,
+                               on_result=on_result)
 
     out = {}
     t_start = time.time()
     for n_done, (cell_idx, config_keys) in enumerate(all_cells):
-        out[config_keys] = evaluate_cell_ref(
-            config_keys, cell_idx, tests=tests, tests_file=tests_file,
-            seed=seed)
+        with trace_span("cell", cell=cell_idx, backend="ref"):
+            out[config_keys] = evaluate_cell_ref(
+                config_keys, cell_idx, tests=tests, tests_file=tests_file,
+                seed=seed)
+        if on_result:
+            on_result(config_keys, out[config_keys])
         if progress:
             progress(n_done + 1, len(all_cells), time.time() - t_start,
                      ", ".join(config_keys))
@@ -131,7 +139,7 @@ def run_scores(tests_file=None, tests=None, backend="auto", cells=None,
 
 
 def _run_scores_hip(all_cells, tests, tests_file, seed, progress,
-                    n_streams=4):
+                    n_streams=4, on_result=None):
     """Device sweep: prebuild the shared caches (views, folds, balanced
     groups) on the default stream, then evaluate cells concurrently on
     worker threads with one HIP stream each — host-side cell bookkeeping
@@ -141,14 +149,16 @@ def _run_scores_hip(all_cells, tests, tests_file, seed, progress,
 
     import torch
 
+    from ..utils.trace import trace_span
     from .hip_cell import SweepContext
 
     context = SweepContext(tests=tests, tests_file=tests_file, seed=seed)
 
-    for cell_idx, config_keys in all_cells:   # warm shared caches
-        context.labels_for(config_keys)
-        context.view_for(config_keys)
-        context.balanced_for(config_keys, cell_idx)
+    with trace_span("prebuild_caches", n_cells=len(all_cells)):
+        for cell_idx, config_keys in all_cells:   # warm shared caches
+            context.labels_for(config_keys)
+            context.view_for(config_keys)
+            context.balanced_for(config_keys, cell_idx)
 
     out = {}
     lock = threading.Lock()
@@ -158,11 +168,14 @@ def _run_scores_hip(all_cells, tests, tests_file, seed, progress,
     def eval_one(args):
         cell_idx, config_keys = args
         stream = torch.cuda.Stream()
-        with torch.cuda.stream(stream):
-            result = context.evaluate_cell(config_keys, cell_idx)
+        with trace_span("cell", cell=cell_idx, backend="hip"):
+            with torch.cuda.stream(stream):
+                result = context.evaluate_cell(config_keys, cell_idx)
         with lock:
             out[config_keys] = result
             n_done[0] += 1
+            if on_result:
+                on_result(config_keys, result)
             if progress:
                 progress(n_done[0], len(all_cells), time.time() - t_start,
                          ", ".join(config_keys))
@@ -186,23 +199,69 @@ def _auto_backend():
     return "ref"
 
 
-def write_scores(tests_file=None, scores_file=SCORES_FILE, backend="auto"):
+def _load_checkpoint(path):
+    """Sequence-of-pickles checkpoint -> {config_keys: value}."""
+    import os
+    done = {}
+    if not os.path.exists(path):
+        return done
+    with open(path, "rb") as fd:
+        while True:
+            try:
+                keys, value = pickle.load(fd)
+                done[keys] = value
+            except EOFError:
+                break
+            except Exception:
+                break  # truncated trailing record from a crash: drop it
+    return done
+
+
+def write_scores(tests_file=None, scores_file=SCORES_FILE, backend="auto",
+                 checkpoint=None):
     """Full 216-cell sweep (sharded across ranks if distributed is
-    initialized) -> scores.pkl on rank 0."""
+    initialized) -> scores.pkl on rank 0.
+
+    checkpoint: optional path PREFIX for crash-restart resumability (the
+    scores-stage analogue of the reference run stage's log.txt,
+    experiment.py:222-237): each completed cell is appended to
+    <prefix>.rank<r>; a re-invocation skips completed cells.
+    """
     from ..parallel import comm
+    from ..utils.trace import trace_span
 
     rank, world = comm.rank_world()
     my_cells = comm.shard_cells(world, rank)
 
-    def progress(done, total, elapsed, name):
-        eta = elapsed / done * (total - done)
-        print(f"[rank {rank}] {done}/{total} {name} "
+    done = {}
+    on_result = None
+    if checkpoint:
+        ckpt_path = f"{checkpoint}.rank{rank}"
+        done = _load_checkpoint(ckpt_path)
+        all_keys = list(iter_config_keys())
+        my_cells = [c for c in my_cells if all_keys[c] not in done]
+        ckpt_fd = open(ckpt_path, "ab")
+        ckpt_lock = __import__("threading").Lock()
+
+        def on_result(keys, value):
+            with ckpt_lock:
+                pickle.dump((keys, value), ckpt_fd)
+                ckpt_fd.flush()
+
+    def progress(done_n, total, elapsed, name):
+        eta = elapsed / done_n * (total - done_n)
+        print(f"[rank {rank}] {done_n}/{total} {name} "
               f"({elapsed:.0f}s elapsed, eta {eta:.0f}s)", flush=True)
 
-    result = run_scores(tests_file=tests_file, backend=backend,
-                        cells=my_cells, progress=progress)
+    with trace_span("scores_sweep", rank_cells=len(my_cells)):
+        result = run_scores(tests_file=tests_file, backend=backend,
+                            cells=my_cells, progress=progress,
+                            on_result=on_result)
+    result.update(done)
     result = comm.gather_scores(result)
 
+    if checkpoint:
+        ckpt_fd.close()
     if rank == 0:
         with open(scores_file, "wb") as fd:
             pickle.dump(result, fd)
