@@ -139,11 +139,12 @@ def run_scores(tests_file=None, tests=None, backend="auto", cells=None,
 
 
 def _run_scores_hip(all_cells, tests, tests_file, seed, progress,
-                    n_streams=4, on_result=None):
+                    n_streams=None, on_result=None):
     """Device sweep: prebuild the shared caches (views, folds, balanced
     groups) on the default stream, then evaluate cells concurrently on
     worker threads with one HIP stream each — host-side cell bookkeeping
     overlaps other cells' kernels (extension calls release the GIL)."""
+    import os
     import threading
     from concurrent.futures import ThreadPoolExecutor
 
@@ -151,6 +152,9 @@ def _run_scores_hip(all_cells, tests, tests_file, seed, progress,
 
     from ..utils.trace import trace_span
     from .hip_cell import SweepContext
+
+    if n_streams is None:
+        n_streams = int(os.environ.get("FLAKE16_STREAMS", "4"))
 
     context = SweepContext(tests=tests, tests_file=tests_file, seed=seed)
 

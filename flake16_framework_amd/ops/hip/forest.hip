@@ -927,28 +927,44 @@ __global__ void small_subtree_kernel(ForestDev a,
                         }
                     }
                 } else {
-                    // iterate occupied code values ascending
-                    int v = cmin;
-                    while (v < cmax) {
-                        const unsigned long long lm =
-                            mask & __ballot(in && my_code <= v);
-                        const long nL = __popcll(lm);
-                        const long n1L = __popcll(lm & lab_mask);
+                    // parallel-rank: every in-mask lane scores its OWN
+                    // code as the threshold candidate (duplicates give
+                    // identical (score, bin) pairs; the argmax tie-break
+                    // to the lowest bin matches the ascending sequential
+                    // scan exactly)
+                    int cnt = 0, cnt1 = 0;
+                    for (int d = 0; d < 64; ++d) {
+                        const int cd = __shfl(my_code, d);
+                        const bool le = ((mask >> d) & 1ULL) &&
+                                        cd <= my_code;
+                        cnt += le;
+                        cnt1 += le && ((lab_mask >> d) & 1ULL);
+                    }
+                    double sc = -1.0e300;
+                    int sb = 0x7FFFFFFF, snl = 0;
+                    if (in && my_code < cmax) {
+                        const long nL = cnt, n1L = cnt1;
                         const long n0L = nL - n1L, nR = n - nL;
                         const long n1R = c1 - n1L, n0R = c0 - n0L;
-                        double sc = (double)(n0L * n0L + n1L * n1L)
-                                        / (double)nL
-                                    + (double)(n0R * n0R + n1R * n1R)
-                                        / (double)nR;
-                        if (sc > best_s) {
-                            best_s = sc; best_f = f; best_b = v;
-                            best_nl = (int)nL;
+                        sc = (double)(n0L * n0L + n1L * n1L) / (double)nL
+                           + (double)(n0R * n0R + n1R * n1R) / (double)nR;
+                        sb = my_code;
+                        snl = (int)nL;
+                    }
+                    for (int d = 32; d > 0; d >>= 1) {
+                        const double os = __shfl_down(sc, d);
+                        const int ob = __shfl_down(sb, d);
+                        const int onl = __shfl_down(snl, d);
+                        if (os > sc || (os == sc && ob < sb)) {
+                            sc = os; sb = ob; snl = onl;
                         }
-                        // next occupied value > v
-                        int nxt = (in && my_code > v) ? my_code : 256;
-                        for (int d = 32; d > 0; d >>= 1)
-                            nxt = min(nxt, __shfl_xor(nxt, d));
-                        v = nxt;
+                    }
+                    sc = __shfl(sc, 0);
+                    sb = __shfl(sb, 0);
+                    snl = __shfl(snl, 0);
+                    if (sc > best_s) {
+                        best_s = sc; best_f = f; best_b = sb;
+                        best_nl = snl;
                     }
                 }
             }

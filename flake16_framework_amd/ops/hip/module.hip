@@ -16,6 +16,7 @@
 
 #include <vector>
 #include <algorithm>
+#include <cstdlib>
 
 #define CHECK_HIP(expr)                                                     \
     do {                                                                    \
@@ -88,7 +89,9 @@ std::vector<at::Tensor> forest_fit(
 
     // Histogram-subtraction pools (see forest.hip): sized for the worst
     // per-level allocation, 2 slots per splitting node >= HIST_SAVE_MIN.
-    const int HIST_SAVE_MIN = 2048;
+    int HIST_SAVE_MIN = 2048;
+    if (const char* e = getenv("FLAKE16_HIST_SAVE_MIN"))
+        HIST_SAVE_MIN = atoi(e);
     const long pool_cap =
         std::min<long>(2 * (S / HIST_SAVE_MIN) + 8, 32768);
     auto hist_pool0 = at::empty({pool_cap * FPAD * 256},
@@ -141,7 +144,8 @@ std::vector<at::Tensor> forest_fit(
     a.small_count = small_count.data_ptr<int>();
     a.small_cap = (int)(S + 2);
 
-    const int GRID = 4096;
+    int GRID = 4096;
+    if (const char* e = getenv("FLAKE16_FIT_GRID")) GRID = atoi(e);
     const int CHUNK = 8;
     int cur = 0;
     long lev = 0;
