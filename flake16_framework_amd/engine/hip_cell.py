@@ -59,6 +59,8 @@ class SweepContext:
     """Caches dataset views shared across cells of one scores sweep."""
 
     def __init__(self, tests=None, tests_file=None, seed=0, device=None):
+        import threading
+
         self.ops = get_ops()
         self.device = device or torch.device("cuda")
         self.tests = tests if tests is not None else load_tests(tests_file)
@@ -67,12 +69,35 @@ class SweepContext:
         self._labels = {}     # flaky_key -> (labels u8, folds, pair tensors)
         self._balanced = {}   # (flaky, fset, prep, bal) -> per-fold data
         self._projects = None
+        # Per-key build locks: worker threads build missing cache entries
+        # concurrently (on their own HIP streams); the builder synchronizes
+        # its stream before publishing so consumers on other streams see
+        # completed tensors.
+        self._meta_lock = threading.Lock()
+        self._key_locks = {}
+
+    def _build_lock(self, key):
+        with self._meta_lock:
+            lock = self._key_locks.get(key)
+            if lock is None:
+                lock = self._key_locks[key] = __import__("threading").Lock()
+        return lock
+
+    def _publish_barrier(self):
+        if self.device.type == "cuda":
+            torch.cuda.current_stream(self.device).synchronize()
 
     # -- labels / folds / prediction index tensors -------------------------
     def labels_for(self, keys):
         flaky_key = keys[0]
         if flaky_key in self._labels:
             return self._labels[flaky_key]
+        with self._build_lock(("labels", flaky_key)):
+            if flaky_key in self._labels:
+                return self._labels[flaky_key]
+            return self._labels_build(keys, flaky_key)
+
+    def _labels_build(self, keys, flaky_key):
         flaky_label, feature_set, *_ = resolve(keys)
         features, labels_b, projects = load_feat_lab_proj(
             flaky_label, feature_set, tests=self.tests)
@@ -97,6 +122,7 @@ class SweepContext:
             "pair_row": torch.from_numpy(pair_row).to(self.device),
             "pair_fold": torch.from_numpy(pair_fold).to(self.device),
         }
+        self._publish_barrier()
         self._labels[flaky_key] = entry
         return entry
 
@@ -106,6 +132,12 @@ class SweepContext:
         ck = (fset_key, prep_key)
         if ck in self._views:
             return self._views[ck]
+        with self._build_lock(("view", ck)):
+            if ck in self._views:
+                return self._views[ck]
+            return self._view_build(keys, ck)
+
+    def _view_build(self, keys, ck):
         ops = self.ops
         flaky_label, feature_set, preproc, *_ = resolve(keys)
         features, _, _ = load_feat_lab_proj(flaky_label, feature_set,
@@ -123,6 +155,7 @@ class SweepContext:
         codes_all = ops.bin_codes(X32, cuts_dev, cut_off_dev, F)
         view = {"X32": X32, "cuts_dev": cuts_dev, "cut_off": cut_off_dev,
                 "codes_all": codes_all, "F": F}
+        self._publish_barrier()
         self._views[ck] = view
         return view
 
@@ -278,11 +311,16 @@ class SweepContext:
     def balanced_for(self, keys, cell_idx):
         """Per-fold balanced training codes for the cell's balance group:
         [(codes_b, yb)] x 10.  k-NN work is batched across folds."""
-        from .scores import job_ids_for
-
         bk = tuple(keys[:4])
         if bk in self._balanced:
             return self._balanced[bk]
+        with self._build_lock(("balanced", bk)):
+            if bk in self._balanced:
+                return self._balanced[bk]
+            return self._balanced_build(keys, cell_idx, bk)
+
+    def _balanced_build(self, keys, cell_idx, bk):
+        from .scores import job_ids_for
 
         ops = self.ops
         _, _, _, balancing, _ = resolve(keys)
@@ -314,6 +352,7 @@ class SweepContext:
             codes_b = ops.bin_codes(Xs[i], view["cuts_dev"], view["cut_off"],
                                     F)
             out.append((codes_b, torch.from_numpy(ys[i]).to(self.device)))
+        self._publish_barrier()
         self._balanced[bk] = out
         return out
 

@@ -156,13 +156,10 @@ def _run_scores_hip(all_cells, tests, tests_file, seed, progress,
     if n_streams is None:
         n_streams = int(os.environ.get("FLAKE16_STREAMS", "4"))
 
+    # Shared caches (views, folds, balanced groups) build lazily under
+    # per-key locks: the first worker to need an entry builds it on its own
+    # stream while other workers proceed with other cells.
     context = SweepContext(tests=tests, tests_file=tests_file, seed=seed)
-
-    with trace_span("prebuild_caches", n_cells=len(all_cells)):
-        for cell_idx, config_keys in all_cells:   # warm shared caches
-            context.labels_for(config_keys)
-            context.view_for(config_keys)
-            context.balanced_for(config_keys, cell_idx)
 
     out = {}
     lock = threading.Lock()
