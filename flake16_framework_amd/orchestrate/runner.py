@@ -84,8 +84,11 @@ def mode_flags(mode, data_file):
     }[mode]
 
 
-def manage_container(cont_name, *commands, subjects_dir=SUBJECTS_DIR,
-                     data_dir=CONT_DATA_DIR, run=sp.run):
+def manage_container(cont_name, *commands, subjects_dir=None,
+                     data_dir=None, run=sp.run):
+    subjects_dir = subjects_dir or os.environ.get("FLAKE16_SUBJECTS_DIR",
+                                                  SUBJECTS_DIR)
+    data_dir = data_dir or os.environ.get("FLAKE16_DATA_DIR", CONT_DATA_DIR)
     """Inside the container: run the subject's pre-commands, then pytest
     with the plugin blacklist, our collectors, and the mode flags."""
     proj, mode, _ = cont_name.split("_", 2)
@@ -118,16 +121,30 @@ def docker_run_argv(cont_name, commands, host_data_dir):
     ]
 
 
+def _local_run_argv(cont_name, commands):
+    """Local (no-docker) mode: the per-run isolation is a fresh python
+    process instead of a container.  Enabled by FLAKE16_LOCAL_RUN=1 with
+    FLAKE16_SUBJECTS_DIR / FLAKE16_DATA_DIR pointing at the work tree —
+    used by the integration tests and available for docker-less hosts."""
+    import sys
+    return [sys.executable, "-m", "flake16_framework_amd.cli", "container",
+            cont_name, *commands]
+
+
 def run_container(args, runner=None, stdout_dir=STDOUT_DIR):
     """Launch one container run; append its stdout; report success."""
     cont_name, commands = args
     host_data_dir = os.path.join(os.getcwd(), DATA_DIR)
     stdout_file = os.path.join(stdout_dir, cont_name)
-    argv = docker_run_argv(cont_name, commands, host_data_dir)
+
+    if os.environ.get("FLAKE16_LOCAL_RUN"):
+        argv = _local_run_argv(cont_name, commands)
+    else:
+        argv = docker_run_argv(cont_name, commands, host_data_dir)
 
     if runner is None:
         def runner(argv, fd):
-            return sp.run(argv, stdout=fd).returncode
+            return sp.run(argv, stdout=fd, stderr=sp.STDOUT).returncode
 
     with open(stdout_file, "a") as fd:
         returncode = runner(argv, fd)
