@@ -52,23 +52,34 @@ def init_from_env(backend=None):
 
 
 def shard_cells(world, rank, n_cells=None):
-    """Cost-balanced static shard of the grid cells: greedy assignment of
-    cells (heaviest first) to the currently lightest rank.  Deterministic.
+    """Cost-balanced static shard of the grid cells, BALANCE-GROUP aware:
+    the 3 model cells of a (flaky x feature-set x preproc x balancing)
+    group share their balanced folds (engine cache), so whole groups are
+    assigned greedily (heaviest first) to the lightest rank — no rank
+    re-balances a group another rank already owns.  Deterministic.
     Returns the sorted list of cell indices owned by `rank`."""
+    from ..configgrid import balance_group_index
+
     keys = list(iter_config_keys())
     if n_cells is not None:
         keys = keys[:n_cells]
-    costs = [cell_cost_estimate(k) for k in keys]
-    order = sorted(range(len(keys)), key=lambda i: (-costs[i], i))
+
+    groups = {}
+    for i, k in enumerate(keys):
+        groups.setdefault(balance_group_index(k), []).append(i)
+    gcost = {g: sum(cell_cost_estimate(keys[i]) for i in cells)
+             for g, cells in groups.items()}
+    order = sorted(groups, key=lambda g: (-gcost[g], g))
 
     load = [0.0] * world
-    owner = [0] * len(keys)
-    for i in order:
+    mine = []
+    for g in order:
         r = min(range(world), key=lambda j: (load[j], j))
-        owner[i] = r
-        load[r] += costs[i]
+        load[r] += gcost[g]
+        if r == rank:
+            mine.extend(groups[g])
 
-    return sorted(i for i in range(len(keys)) if owner[i] == rank)
+    return sorted(mine)
 
 
 def _pack(result, cell_order, projects):
