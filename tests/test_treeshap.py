@@ -42,8 +42,6 @@ class TestTreeShapRef:
         forest, codes = _fit_small(n=120, f=8, seed=3, n_trees=5,
                                    bootstrap=True)
         shap = forest_shap(forest, codes[:20], 8)
-        for tree in forest.trees:
-            pass
         # model output: mean class-0 prob; base: mean over root covers
         out = np.zeros(20)
         base = 0.0
