@@ -793,6 +793,293 @@ __global__ void et_split_kernel(ForestDev a) {
 }
 
 // ---------------------------------------------------------------------------
+// Candidate-only Random-Forest level kernel.
+//
+// RF evaluates max_features (= sqrt(F) ~ 4) candidate features per node,
+// but the generic histogram kernel accumulates all 16.  Here pass 1
+// computes per-feature min/max (constancy) + class counts; the Fisher-
+// Yates walk then fixes the <= max_features candidates, and pass 2
+// histograms ONLY those (~4 LDS atomics per sample instead of 16).
+// Split scores, tie-breaks and partition are identical to
+// hist_split_kernel, so trees are unchanged.  No histogram-subtraction
+// pools on this path (children of other candidates are useless to them).
+// ---------------------------------------------------------------------------
+__launch_bounds__(HBLK)
+__global__ void rf_cand_split_kernel(ForestDev a) {
+    __shared__ uint32_t hist[FPAD * 256];   // rows = candidate index
+    __shared__ int sh_scan[HBLK];
+    __shared__ int sh_min[FPAD], sh_max[FPAD];
+    __shared__ int sh_cand[FPAD], sh_ncand;
+    __shared__ double sh_score[FPAD];
+    __shared__ int sh_bin[FPAD], sh_nL[FPAD];
+    __shared__ int sh_bestf, sh_bestbin, sh_bestnL;
+    __shared__ int sh_loff, sh_roff, sh_lid;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int n_items = *a.cur_count;
+
+    for (int wi = blockIdx.x; wi < n_items; wi += gridDim.x) {
+        WorkItem it = a.cur[wi];
+        const int n = it.end - it.start;
+        const long sbase = a.j_sidx_off[it.job];
+        const long nbase = a.j_node_off[it.job];
+        const uint32_t key = (uint32_t)a.j_key[it.job];
+        const int F = a.F;
+
+        if (tid < F) {
+            sh_min[tid] = 256;
+            sh_max[tid] = -1;
+        }
+        __syncthreads();
+
+        // Pass 1: per-feature min/max + class-1 count.
+        int lmin[FPAD], lmax[FPAD];
+        #pragma unroll
+        for (int f = 0; f < FPAD; ++f) { lmin[f] = 256; lmax[f] = -1; }
+        int lc1 = 0;
+        for (int i = it.start + tid; i < it.end; i += HBLK) {
+            int row = a.sidx_cur[sbase + i];
+            uint4 cw = *reinterpret_cast<const uint4*>(
+                a.codes + (size_t)row * FPAD);
+            uint32_t w[4] = {cw.x, cw.y, cw.z, cw.w};
+            lc1 += a.labels[row];
+            for (int f = 0; f < F; ++f) {
+                const int b = (int)((w[f >> 2] >> ((f & 3) * 8)) & 0xFFu);
+                lmin[f] = min(lmin[f], b);
+                lmax[f] = max(lmax[f], b);
+            }
+        }
+        for (int d = 32; d > 0; d >>= 1) lc1 += __shfl_down(lc1, d);
+        if (lane == 0) sh_scan[wave] = lc1;
+        for (int f = 0; f < F; ++f) {
+            int mn = lmin[f], mx = lmax[f];
+            for (int d = 32; d > 0; d >>= 1) {
+                mn = min(mn, __shfl_xor(mn, d));
+                mx = max(mx, __shfl_xor(mx, d));
+            }
+            if (lane == 0) {
+                atomicMin(&sh_min[f], mn);
+                atomicMax(&sh_max[f], mx);
+            }
+        }
+        __syncthreads();
+        const int c1 = sh_scan[0] + sh_scan[1] + sh_scan[2] + sh_scan[3];
+        const int c0 = n - c1;
+        __syncthreads();
+
+        if (tid == 0) {
+            a.ncnt0[nbase + it.node] = (float)c0;
+            a.ncnt1[nbase + it.node] = (float)c1;
+        }
+        if (n < 2 || c0 == 0 || c1 == 0) {
+            __syncthreads();
+            continue;
+        }
+
+        // Candidate walk (thread 0): same permutation/constancy semantics
+        // as the histogram path's phase 4.
+        if (tid == 0) {
+            int perm[FPAD];
+            for (int f = 0; f < F; ++f) perm[f] = f;
+            uint32_t tag = TAG_FEATSEL | ((uint32_t)(it.depth & 0xFF) << 8);
+            for (int i = 0; i < F - 1; ++i) {
+                uint32_t u = philox_draw(tag, (uint32_t)it.start,
+                                         (uint32_t)it.end, (uint32_t)i,
+                                         a.seed, key);
+                int j = i + (int)philox_bounded(u, (uint32_t)(F - i));
+                int t = perm[i]; perm[i] = perm[j]; perm[j] = t;
+            }
+            int nc = 0;
+            for (int i = 0; i < F && nc < a.max_features; ++i) {
+                int f = perm[i];
+                if (sh_min[f] != sh_max[f]) sh_cand[nc++] = f;
+            }
+            sh_ncand = nc;
+        }
+        __syncthreads();
+
+        const int ncand = sh_ncand;
+        if (ncand == 0) {
+            __syncthreads();
+            continue;
+        }
+
+        // Pass 2: packed histograms of the candidate features only.
+        for (int i = tid; i < ncand * 256; i += HBLK) hist[i] = 0;
+        __syncthreads();
+        for (int i = it.start + tid; i < it.end; i += HBLK) {
+            int row = a.sidx_cur[sbase + i];
+            uint4 cw = *reinterpret_cast<const uint4*>(
+                a.codes + (size_t)row * FPAD);
+            uint32_t w[4] = {cw.x, cw.y, cw.z, cw.w};
+            const uint32_t inc = 1u | ((uint32_t)a.labels[row] << 16);
+            for (int ci = 0; ci < ncand; ++ci) {
+                const int f = sh_cand[ci];
+                uint32_t b = (w[f >> 2] >> ((f & 3) * 8)) & 0xFFu;
+                atomicAdd(&hist[ci * 256 + b], inc);
+            }
+        }
+        __syncthreads();
+
+        // Evaluate: wave w handles candidates w, w+4, ... (identical fp64
+        // scores and tie-breaks to hist_split_kernel phase 5).
+        for (int ci = wave; ci < ncand; ci += HBLK / 64) {
+            const int f = sh_cand[ci];
+            const int bmin = sh_min[f], bmax = sh_max[f];
+            int ln[4], l1[4];
+            int tn = 0, t1 = 0;
+            for (int k = 0; k < 4; ++k) {
+                uint32_t v = hist[ci * 256 + lane * 4 + k];
+                ln[k] = (int)(v & 0xFFFFu);
+                l1[k] = (int)(v >> 16);
+                tn += ln[k];
+                t1 += l1[k];
+            }
+            int scn = tn, sc1 = t1;
+            for (int d = 1; d < 64; d <<= 1) {
+                int un = __shfl_up(scn, d);
+                int u1 = __shfl_up(sc1, d);
+                if (lane >= d) { scn += un; sc1 += u1; }
+            }
+            const int excl_n = scn - tn, excl_1 = sc1 - t1;
+
+            double best_s = -1.0;
+            int best_b = -1, best_nl = 0;
+            int cn = excl_n, c1f = excl_1;
+            for (int k = 0; k < 4; ++k) {
+                int b = lane * 4 + k;
+                cn += ln[k];
+                c1f += l1[k];
+                if (ln[k] == 0 || b < bmin || b >= bmax) continue;
+                long nL = cn, n1L = c1f;
+                long n0L = nL - n1L, nR = n - nL;
+                long n1R = c1 - n1L, n0R = c0 - n0L;
+                double s = (double)(n0L * n0L + n1L * n1L) / (double)nL
+                         + (double)(n0R * n0R + n1R * n1R) / (double)nR;
+                if (s > best_s) { best_s = s; best_b = b; best_nl = (int)nL; }
+            }
+            for (int d = 32; d > 0; d >>= 1) {
+                double os = __shfl_down(best_s, d);
+                int ob = __shfl_down(best_b, d);
+                int onl = __shfl_down(best_nl, d);
+                if (os > best_s || (os == best_s && ob != -1 &&
+                                    (best_b == -1 || ob < best_b))) {
+                    best_s = os; best_b = ob; best_nl = onl;
+                }
+            }
+            if (lane == 0) {
+                sh_score[ci] = best_s;
+                sh_bin[ci] = best_b;
+                sh_nL[ci] = best_nl;
+            }
+        }
+        __syncthreads();
+
+        if (tid == 0) {
+            double best_s = -1.0e300;
+            int bf = -1, bb = -1, bnl = 0;
+            for (int ci = 0; ci < ncand; ++ci) {
+                if (sh_bin[ci] >= 0 && sh_score[ci] > best_s) {
+                    best_s = sh_score[ci];
+                    bf = sh_cand[ci];
+                    bb = sh_bin[ci];
+                    bnl = sh_nL[ci];
+                }
+            }
+            sh_bestf = bf;
+            sh_bestbin = bb;
+            sh_bestnL = bnl;
+            if (bf >= 0) {
+                int l = atomicAdd(&a.node_alloc[it.job], 2);
+                a.nfeat[nbase + it.node] = bf;
+                a.nsplit[nbase + it.node] = bb;
+                a.nleft[nbase + it.node] = l;
+                sh_lid = l;
+            }
+            sh_loff = 0;
+            sh_roff = 0;
+        }
+        __syncthreads();
+
+        const int bf = sh_bestf;
+        if (bf < 0) {
+            __syncthreads();
+            continue;
+        }
+        const int bb = sh_bestbin;
+        const int nL = sh_bestnL;
+
+        // Partition (identical to hist_split_kernel phase 7).
+        for (int base = it.start; base < it.end; base += HBLK) {
+            const int i = base + tid;
+            const bool valid = i < it.end;
+            int row = 0, flag = 0;
+            if (valid) {
+                row = a.sidx_cur[sbase + i];
+                uint32_t b = a.codes[(size_t)row * FPAD + bf];
+                flag = (int)(b <= (uint32_t)bb);
+            }
+            const unsigned long long lm = __ballot(valid && flag);
+            const unsigned long long below = (1ULL << lane) - 1ULL;
+            const int left_rank = __popcll(lm & below);
+            if (lane == 0) sh_scan[wave] = __popcll(lm);
+            __syncthreads();
+            int wave_left_excl = 0;
+            for (int ww = 0; ww < wave; ++ww)
+                wave_left_excl += sh_scan[ww];
+            const int tile_left = sh_scan[0] + sh_scan[1] + sh_scan[2]
+                                  + sh_scan[3];
+            const int tile_n = min(HBLK, it.end - base);
+            if (valid) {
+                if (flag)
+                    a.sidx_nxt[sbase + it.start + sh_loff + wave_left_excl
+                               + left_rank] = row;
+                else
+                    a.sidx_nxt[sbase + it.start + nL + sh_roff
+                               + (i - base) - (wave_left_excl + left_rank)]
+                        = row;
+            }
+            __syncthreads();
+            if (tid == 0) {
+                sh_loff += tile_left;
+                sh_roff += tile_n - tile_left;
+            }
+            __syncthreads();
+        }
+
+        if (tid == 0) {
+            WorkItem left = {it.job, sh_lid, it.start, it.start + nL,
+                             it.depth + 1, -1};
+            WorkItem right = {it.job, sh_lid + 1, it.start + nL, it.end,
+                              it.depth + 1, -1};
+            WorkItem* q[2];
+            int qn = 0, sn = 0;
+            WorkItem* sq[2];
+            if (nL <= SMALL_N) sq[sn++] = &left; else q[qn++] = &left;
+            if (it.end - it.start - nL <= SMALL_N) sq[sn++] = &right;
+            else q[qn++] = &right;
+            if (qn) {
+                const int qi = atomicAdd(a.nxt_count, qn);
+                if (qi + qn <= a.work_cap)
+                    for (int x = 0; x < qn; ++x) a.nxt[qi + x] = *q[x];
+                else
+                    atomicExch(a.err_flag, 1);
+            }
+            if (sn) {
+                const int si = atomicAdd(a.small_count, sn);
+                if (si + sn <= a.small_cap)
+                    for (int x = 0; x < sn; ++x) a.small[si + x] = *sq[x];
+                else
+                    atomicExch(a.err_flag, 1);
+            }
+        }
+        __syncthreads();
+    }
+}
+
+// ---------------------------------------------------------------------------
 // Wave-per-subtree builder for nodes with n <= SMALL_N (=64) samples.
 //
 // One 64-lane wave finishes the WHOLE subtree: each lane holds one
