@@ -130,6 +130,7 @@ __global__ void hist_split_kernel(ForestDev a) {
     __shared__ int sh_bestf, sh_bestbin, sh_bestnL;
     __shared__ int sh_loff, sh_roff;
     __shared__ int sh_lid;                 // allocated left-child node id
+    __shared__ uint32_t sh_draws[FPAD];
     __shared__ int sh_accum_small;         // phase-8 plan flags
     __shared__ int sh_slot_small, sh_slot_large, sh_small_is_left;
 
@@ -222,18 +223,22 @@ __global__ void hist_split_kernel(ForestDev a) {
         }
         __syncthreads();
 
-        // Phase 4 (thread 0): feature permutation (partial Fisher-Yates,
-        // counters (TAG_FEATSEL|depth<<8, start, end, i)) and the candidate
-        // list: walk perm, non-constant features until max_features.
+        // Phase 4: feature permutation (partial Fisher-Yates, counters
+        // (TAG_FEATSEL|depth<<8, start, end, i)) — draws computed by the
+        // first F-1 threads in parallel, swap walk on thread 0 — and the
+        // candidate list: walk perm, non-constant until max_features.
+        if (tid < FPAD)
+            sh_draws[tid] = philox_draw(
+                TAG_FEATSEL | ((uint32_t)(it.depth & 0xFF) << 8),
+                (uint32_t)it.start, (uint32_t)it.end, (uint32_t)tid,
+                a.seed, key);
+        __syncthreads();
         if (tid == 0) {
             int perm[FPAD];
             for (int f = 0; f < F; ++f) perm[f] = f;
-            uint32_t tag = TAG_FEATSEL | ((uint32_t)(it.depth & 0xFF) << 8);
             for (int i = 0; i < F - 1; ++i) {
-                uint32_t u = philox_draw(tag, (uint32_t)it.start,
-                                         (uint32_t)it.end, (uint32_t)i,
-                                         a.seed, key);
-                int j = i + (int)philox_bounded(u, (uint32_t)(F - i));
+                int j = i + (int)philox_bounded(sh_draws[i],
+                                                (uint32_t)(F - i));
                 int t = perm[i]; perm[i] = perm[j]; perm[j] = t;
             }
             int nc = 0;
@@ -547,6 +552,7 @@ __global__ void et_split_kernel(ForestDev a) {
     __shared__ int sh_scan[HBLK];
     __shared__ int sh_min[FPAD], sh_max[FPAD];
     __shared__ int sh_cand[FPAD], sh_cbin[FPAD], sh_ncand;
+    __shared__ uint32_t sh_draws[FPAD], sh_tdraws[FPAD];
     __shared__ int sh_cnt[4][FPAD], sh_cnt1[4][FPAD];
     __shared__ int sh_bestf, sh_bestbin, sh_bestnL;
     __shared__ int sh_loff, sh_roff, sh_lid;
@@ -614,29 +620,34 @@ __global__ void et_split_kernel(ForestDev a) {
             continue;
         }
 
-        // Permutation walk + threshold draws (thread 0).
+        // Permutation walk + threshold draws: all Philox draws computed
+        // by the first F threads in parallel, walk on thread 0.
+        if (tid < FPAD) {
+            sh_draws[tid] = philox_draw(
+                TAG_FEATSEL | ((uint32_t)(it.depth & 0xFF) << 8),
+                (uint32_t)it.start, (uint32_t)it.end, (uint32_t)tid,
+                a.seed, key);
+            sh_tdraws[tid] = philox_draw(
+                TAG_THRESH | ((uint32_t)(it.depth & 0xFF) << 8),
+                (uint32_t)it.start, (uint32_t)it.end, (uint32_t)tid,
+                a.seed, key);
+        }
+        __syncthreads();
         if (tid == 0) {
             int perm[FPAD];
             for (int f = 0; f < F; ++f) perm[f] = f;
-            uint32_t tag = TAG_FEATSEL | ((uint32_t)(it.depth & 0xFF) << 8);
             for (int i = 0; i < F - 1; ++i) {
-                uint32_t u = philox_draw(tag, (uint32_t)it.start,
-                                         (uint32_t)it.end, (uint32_t)i,
-                                         a.seed, key);
-                int j = i + (int)philox_bounded(u, (uint32_t)(F - i));
+                int j = i + (int)philox_bounded(sh_draws[i],
+                                                (uint32_t)(F - i));
                 int t = perm[i]; perm[i] = perm[j]; perm[j] = t;
             }
-            uint32_t ttag = TAG_THRESH | ((uint32_t)(it.depth & 0xFF) << 8);
             int nc = 0;
             for (int i = 0; i < F && nc < a.max_features; ++i) {
                 int f = perm[i];
                 if (sh_min[f] == sh_max[f]) continue;
-                uint32_t u = philox_draw(ttag, (uint32_t)it.start,
-                                         (uint32_t)it.end, (uint32_t)f,
-                                         a.seed, key);
                 sh_cand[nc] = f;
                 sh_cbin[nc] = sh_min[f] + (int)philox_bounded(
-                    u, (uint32_t)(sh_max[f] - sh_min[f]));
+                    sh_tdraws[f], (uint32_t)(sh_max[f] - sh_min[f]));
                 ++nc;
             }
             sh_ncand = nc;
@@ -810,6 +821,7 @@ __global__ void rf_cand_split_kernel(ForestDev a) {
     __shared__ int sh_scan[HBLK];
     __shared__ int sh_min[FPAD], sh_max[FPAD];
     __shared__ int sh_cand[FPAD], sh_ncand;
+    __shared__ uint32_t sh_draws[FPAD];
     __shared__ double sh_score[FPAD];
     __shared__ int sh_bin[FPAD], sh_nL[FPAD];
     __shared__ int sh_bestf, sh_bestbin, sh_bestnL;
@@ -878,17 +890,20 @@ __global__ void rf_cand_split_kernel(ForestDev a) {
             continue;
         }
 
-        // Candidate walk (thread 0): same permutation/constancy semantics
-        // as the histogram path's phase 4.
+        // Candidate walk: parallel draws, thread-0 swap walk (same
+        // permutation/constancy semantics as the histogram path).
+        if (tid < FPAD)
+            sh_draws[tid] = philox_draw(
+                TAG_FEATSEL | ((uint32_t)(it.depth & 0xFF) << 8),
+                (uint32_t)it.start, (uint32_t)it.end, (uint32_t)tid,
+                a.seed, key);
+        __syncthreads();
         if (tid == 0) {
             int perm[FPAD];
             for (int f = 0; f < F; ++f) perm[f] = f;
-            uint32_t tag = TAG_FEATSEL | ((uint32_t)(it.depth & 0xFF) << 8);
             for (int i = 0; i < F - 1; ++i) {
-                uint32_t u = philox_draw(tag, (uint32_t)it.start,
-                                         (uint32_t)it.end, (uint32_t)i,
-                                         a.seed, key);
-                int j = i + (int)philox_bounded(u, (uint32_t)(F - i));
+                int j = i + (int)philox_bounded(sh_draws[i],
+                                                (uint32_t)(F - i));
                 int t = perm[i]; perm[i] = perm[j]; perm[j] = t;
             }
             int nc = 0;
@@ -1156,19 +1171,36 @@ __global__ void small_subtree_kernel(ForestDev a,
             }
             if (n < 2 || c0 == 0 || c1 == 0) continue;
 
-            // feature permutation (lane 0, broadcast via LDS)
-            if (lane == 0) {
-                for (int f = 0; f < F; ++f) perm[f] = f;
+            // feature permutation: each of the first F-1 lanes computes
+            // its Philox draw in parallel; lane 0 runs the swap loop with
+            // the draws broadcast by shuffle
+            {
                 uint32_t tag = TAG_FEATSEL |
                                ((uint32_t)(depth & 0xFF) << 8);
+                const uint32_t u_mine = philox_draw(
+                    tag, (uint32_t)s, (uint32_t)e,
+                    (uint32_t)(lane < FPAD ? lane : 0), a.seed, key);
+                if (lane == 0)
+                    for (int f = 0; f < F; ++f) perm[f] = f;
                 for (int i = 0; i < F - 1; ++i) {
-                    uint32_t u = philox_draw(tag, (uint32_t)s, (uint32_t)e,
-                                             (uint32_t)i, a.seed, key);
-                    int j = i + (int)philox_bounded(u, (uint32_t)(F - i));
-                    int t = perm[i]; perm[i] = perm[j]; perm[j] = t;
+                    const uint32_t u = __shfl(u_mine, i);
+                    if (lane == 0) {
+                        int j = i + (int)philox_bounded(
+                            u, (uint32_t)(F - i));
+                        int t = perm[i]; perm[i] = perm[j]; perm[j] = t;
+                    }
                 }
             }
             // lane 0's LDS writes are visible to the wave in program order
+
+            uint32_t u_thresh = 0;
+            if (a.splitter_random) {
+                uint32_t ttag = TAG_THRESH |
+                                ((uint32_t)(depth & 0xFF) << 8);
+                u_thresh = philox_draw(ttag, (uint32_t)s, (uint32_t)e,
+                                       (uint32_t)(lane < FPAD ? lane : 0),
+                                       a.seed, key);
+            }
 
             double best_s = -1.0e300;
             int best_f = -1, best_b = -1, best_nl = 0;
@@ -1191,12 +1223,8 @@ __global__ void small_subtree_kernel(ForestDev a,
                 ++n_eval;
 
                 if (a.splitter_random) {
-                    uint32_t tag = TAG_THRESH |
-                                   ((uint32_t)(depth & 0xFF) << 8);
-                    uint32_t u = philox_draw(tag, (uint32_t)s, (uint32_t)e,
-                                             (uint32_t)f, a.seed, key);
                     const int b = cmin + (int)philox_bounded(
-                        u, (uint32_t)(cmax - cmin));
+                        __shfl(u_thresh, f), (uint32_t)(cmax - cmin));
                     const unsigned long long lm =
                         mask & __ballot(in && my_code <= b);
                     const long nL = __popcll(lm);
