@@ -86,11 +86,13 @@ def mode_flags(mode, data_file):
 
 def manage_container(cont_name, *commands, subjects_dir=None,
                      data_dir=None, run=sp.run):
+    """Inside the container: run the subject's pre-commands, then pytest
+    with the plugin blacklist, our collectors, and the mode flags.
+    Directory overrides via FLAKE16_SUBJECTS_DIR / FLAKE16_DATA_DIR
+    support the local (docker-less) mode."""
     subjects_dir = subjects_dir or os.environ.get("FLAKE16_SUBJECTS_DIR",
                                                   SUBJECTS_DIR)
     data_dir = data_dir or os.environ.get("FLAKE16_DATA_DIR", CONT_DATA_DIR)
-    """Inside the container: run the subject's pre-commands, then pytest
-    with the plugin blacklist, our collectors, and the mode flags."""
     proj, mode, _ = cont_name.split("_", 2)
     proj_dir = os.path.join(subjects_dir, proj, proj)
     data_file = os.path.join(data_dir, cont_name)

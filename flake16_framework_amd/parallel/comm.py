@@ -134,7 +134,9 @@ def gather_scores(result):
     # principle own zero cells — exchange via all_gather_object.
     plists = [None] * world
     d.all_gather_object(plists, projects)
-    projects = next(p for p in plists if p is not None)
+    projects = next((p for p in plists if p is not None), None)
+    if projects is None:
+        return result  # no rank evaluated anything
 
     buf = _pack(result, cell_order, projects)
     use_cuda = d.get_backend() == "nccl"
