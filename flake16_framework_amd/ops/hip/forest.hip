@@ -74,9 +74,36 @@ struct ForestDev {
     WorkItem* __restrict__ small;
     int* __restrict__ small_count;
     int small_cap;
+    // Mid-subtree routing: children with SMALL_N < n <= MID_N are staged
+    // into LDS by mid_subtree_kernel, which recurses internally (no level
+    // round-trips) and farms its <=SMALL_N descendants to the small queue.
+    WorkItem* __restrict__ mid;
+    int* __restrict__ mid_count;
+    int mid_cap;
 };
 
 #define SMALL_N 64
+#define MID_N 1024
+
+// Three-way child routing: <=SMALL_N -> wave-subtree queue, <=MID_N ->
+// LDS-staged mid-subtree queue, else next level.  Called by one thread.
+__device__ __forceinline__ void route_child(const ForestDev& a,
+                                            const WorkItem& w) {
+    const int n = w.end - w.start;
+    if (n <= SMALL_N) {
+        const int i = atomicAdd(a.small_count, 1);
+        if (i < a.small_cap) a.small[i] = w;
+        else atomicExch(a.err_flag, 1);
+    } else if (n <= MID_N && w.hist_slot < 0) {
+        const int i = atomicAdd(a.mid_count, 1);
+        if (i < a.mid_cap) a.mid[i] = w;
+        else atomicExch(a.err_flag, 1);
+    } else {
+        const int i = atomicAdd(a.nxt_count, 1);
+        if (i < a.work_cap) a.nxt[i] = w;
+        else atomicExch(a.err_flag, 1);
+    }
+}
 
 // ---------------------------------------------------------------------------
 // Init: fill per-job sample indices (bootstrap or identity) and root items.
@@ -458,33 +485,10 @@ __global__ void hist_split_kernel(ForestDev a) {
 
             const int sl = sh_small_is_left ? sh_slot_small : sh_slot_large;
             const int sr = sh_small_is_left ? sh_slot_large : sh_slot_small;
-            WorkItem left = {it.job, sh_lid, it.start, it.start + nL,
-                             it.depth + 1, sh_accum_small ? sl : -1};
-            WorkItem right = {it.job, sh_lid + 1, it.start + nL, it.end,
-                              it.depth + 1, sh_accum_small ? sr : -1};
-            // children small enough for the wave-subtree path skip the
-            // level queue entirely
-            WorkItem* q[2];
-            int qn = 0, sn = 0;
-            WorkItem* sq[2];
-            if (nL <= SMALL_N) sq[sn++] = &left; else q[qn++] = &left;
-            if (nR <= SMALL_N) sq[sn++] = &right; else q[qn++] = &right;
-            if (qn) {
-                const int qi = atomicAdd(a.nxt_count, qn);
-                if (qi + qn <= a.work_cap) {
-                    for (int x = 0; x < qn; ++x) a.nxt[qi + x] = *q[x];
-                } else {
-                    atomicExch(a.err_flag, 1);
-                }
-            }
-            if (sn) {
-                const int si = atomicAdd(a.small_count, sn);
-                if (si + sn <= a.small_cap) {
-                    for (int x = 0; x < sn; ++x) a.small[si + x] = *sq[x];
-                } else {
-                    atomicExch(a.err_flag, 1);
-                }
-            }
+            route_child(a, {it.job, sh_lid, it.start, it.start + nL,
+                            it.depth + 1, sh_accum_small ? sl : -1});
+            route_child(a, {it.job, sh_lid + 1, it.start + nL, it.end,
+                            it.depth + 1, sh_accum_small ? sr : -1});
         }
         __syncthreads();
 
@@ -774,30 +778,10 @@ __global__ void et_split_kernel(ForestDev a) {
 
         // Push children (no histogram slots on the ET path).
         if (tid == 0) {
-            WorkItem left = {it.job, sh_lid, it.start, it.start + nL,
-                             it.depth + 1, -1};
-            WorkItem right = {it.job, sh_lid + 1, it.start + nL, it.end,
-                              it.depth + 1, -1};
-            WorkItem* q[2];
-            int qn = 0, sn = 0;
-            WorkItem* sq[2];
-            if (nL <= SMALL_N) sq[sn++] = &left; else q[qn++] = &left;
-            if (it.end - it.start - nL <= SMALL_N) sq[sn++] = &right;
-            else q[qn++] = &right;
-            if (qn) {
-                const int qi = atomicAdd(a.nxt_count, qn);
-                if (qi + qn <= a.work_cap)
-                    for (int x = 0; x < qn; ++x) a.nxt[qi + x] = *q[x];
-                else
-                    atomicExch(a.err_flag, 1);
-            }
-            if (sn) {
-                const int si = atomicAdd(a.small_count, sn);
-                if (si + sn <= a.small_cap)
-                    for (int x = 0; x < sn; ++x) a.small[si + x] = *sq[x];
-                else
-                    atomicExch(a.err_flag, 1);
-            }
+            route_child(a, {it.job, sh_lid, it.start, it.start + nL,
+                            it.depth + 1, -1});
+            route_child(a, {it.job, sh_lid + 1, it.start + nL, it.end,
+                            it.depth + 1, -1});
         }
         __syncthreads();
     }
@@ -1065,31 +1049,391 @@ __global__ void rf_cand_split_kernel(ForestDev a) {
         }
 
         if (tid == 0) {
-            WorkItem left = {it.job, sh_lid, it.start, it.start + nL,
-                             it.depth + 1, -1};
-            WorkItem right = {it.job, sh_lid + 1, it.start + nL, it.end,
-                              it.depth + 1, -1};
-            WorkItem* q[2];
-            int qn = 0, sn = 0;
-            WorkItem* sq[2];
-            if (nL <= SMALL_N) sq[sn++] = &left; else q[qn++] = &left;
-            if (it.end - it.start - nL <= SMALL_N) sq[sn++] = &right;
-            else q[qn++] = &right;
-            if (qn) {
-                const int qi = atomicAdd(a.nxt_count, qn);
-                if (qi + qn <= a.work_cap)
-                    for (int x = 0; x < qn; ++x) a.nxt[qi + x] = *q[x];
-                else
-                    atomicExch(a.err_flag, 1);
-            }
-            if (sn) {
-                const int si = atomicAdd(a.small_count, sn);
-                if (si + sn <= a.small_cap)
-                    for (int x = 0; x < sn; ++x) a.small[si + x] = *sq[x];
-                else
-                    atomicExch(a.err_flag, 1);
-            }
+            route_child(a, {it.job, sh_lid, it.start, it.start + nL,
+                            it.depth + 1, -1});
+            route_child(a, {it.job, sh_lid + 1, it.start + nL, it.end,
+                            it.depth + 1, -1});
         }
+        __syncthreads();
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Mid-subtree builder: one BLOCK finishes the whole subtree of a node with
+// SMALL_N < n <= MID_N samples from LDS-resident data.
+//
+// The node's code rows + labels are staged into LDS ONCE; the block then
+// recurses (smaller-child-first DFS, stack in LDS) over its > SMALL_N
+// descendants with LDS histograms and LDS index partitions — no level
+// round-trips, no global gathers per level (PMC showed the level kernels
+// 40-69% wait-bound on exactly those).  <= SMALL_N descendants are written
+// back (final index arrangement -> sidx buffer) and pushed to the small
+// queue for the wave kernel.  Split semantics, RNG counters (global
+// job-relative (start,end) ranges) and fp64 scores are identical to
+// hist_split_kernel — trees are bit-identical.
+// Handles both splitters (best: prefix scan over bins; random: drawn bin).
+// ---------------------------------------------------------------------------
+struct MidFrame {
+    short s, e;     // LOCAL sample range within the staged window
+    int depth;
+    int node;
+};
+
+#define MID_STACK 16   // smaller-child-first: <= log2(MID_N)+2
+
+__launch_bounds__(HBLK)
+__global__ void mid_subtree_kernel(ForestDev a,
+                                   const int* __restrict__ sidx_level) {
+    __shared__ uint4 m_codes[MID_N];          // 16 KiB staged code rows
+    __shared__ int m_rows[MID_N];             // 4 KiB original global rows
+    __shared__ uint8_t m_lab[MID_N];          // 1 KiB labels
+    __shared__ uint16_t m_idx[MID_N];         // 2 KiB sample ordering
+    __shared__ uint16_t m_idx2[MID_N];        // 2 KiB partition scratch
+    __shared__ uint32_t hist[FPAD * 256];     // 16 KiB packed histogram
+    __shared__ int sh_scan[HBLK];
+    __shared__ int sh_bmin[FPAD], sh_bmax[FPAD];
+    __shared__ int sh_cand[FPAD], sh_ncand;
+    __shared__ uint32_t sh_draws[FPAD];
+    __shared__ double sh_score[FPAD];
+    __shared__ int sh_bin[FPAD], sh_nL[FPAD];
+    __shared__ int sh_bestf, sh_bestbin, sh_bestnL, sh_lid;
+    __shared__ MidFrame stack[MID_STACK];
+    __shared__ int sh_sp;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int n_items = *a.mid_count;
+
+    for (int wi = blockIdx.x; wi < n_items; wi += gridDim.x) {
+        const WorkItem it = a.mid[wi];
+        const int n0 = it.end - it.start;
+        const long sbase = a.j_sidx_off[it.job];
+        const long nbase = a.j_node_off[it.job];
+        const uint32_t key = (uint32_t)a.j_key[it.job];
+        const int F = a.F;
+
+        // Stage the subtree's samples into LDS (one gather, reused by
+        // every internal level).
+        for (int i = tid; i < n0; i += HBLK) {
+            const int row = sidx_level[sbase + it.start + i];
+            m_rows[i] = row;
+            m_codes[i] = *reinterpret_cast<const uint4*>(
+                a.codes + (size_t)row * FPAD);
+            m_lab[i] = a.labels[row];
+            m_idx[i] = (uint16_t)i;
+        }
+        if (tid == 0) {
+            sh_sp = 0;
+            stack[0] = {0, (short)n0, it.depth, it.node};
+        }
+        __syncthreads();
+
+        while (true) {
+            const int sp = sh_sp;
+            if (sp < 0) break;
+            const int ls = stack[sp].s;
+            const int le = stack[sp].e;
+            const int depth = stack[sp].depth;
+            const int node = stack[sp].node;
+            const int n = le - ls;
+            // global job-relative range (the RNG identity)
+            const int gs = it.start + ls;
+            const int ge = it.start + le;
+            __syncthreads();
+            if (tid == 0) --sh_sp;
+
+            // histogram from LDS-resident samples
+            for (int i = tid; i < F * 64; i += HBLK)
+                reinterpret_cast<uint4*>(hist)[i] = uint4{0, 0, 0, 0};
+            __syncthreads();
+            for (int i = ls + tid; i < le; i += HBLK) {
+                const int o = m_idx[i];
+                const uint4 cw = m_codes[o];
+                const uint32_t w[4] = {cw.x, cw.y, cw.z, cw.w};
+                const uint32_t inc = 1u | ((uint32_t)m_lab[o] << 16);
+                for (int f = 0; f < F; ++f) {
+                    uint32_t b = (w[f >> 2] >> ((f & 3) * 8)) & 0xFFu;
+                    atomicAdd(&hist[f * 256 + b], inc);
+                }
+            }
+            __syncthreads();
+
+            // class counts (wave reduce over feature-0 histogram)
+            {
+                int v = (int)(hist[tid] >> 16);
+                for (int d = 32; d > 0; d >>= 1) v += __shfl_down(v, d);
+                if (lane == 0) sh_scan[wave] = v;
+            }
+            __syncthreads();
+            const int c1 = sh_scan[0] + sh_scan[1] + sh_scan[2] + sh_scan[3];
+            const int c0 = n - c1;
+            __syncthreads();
+
+            if (tid == 0) {
+                a.ncnt0[nbase + node] = (float)c0;
+                a.ncnt1[nbase + node] = (float)c1;
+            }
+            if (n < 2 || c0 == 0 || c1 == 0) {
+                __syncthreads();
+                continue;
+            }
+
+            // occupied-bin range per feature
+            if (tid < F) {
+                sh_bmin[tid] = 256;
+                sh_bmax[tid] = -1;
+            }
+            __syncthreads();
+            {
+                const int f3 = tid >> 4;
+                const int seg = (tid & 15) * 16;
+                if (f3 < F) {
+                    int lo = 256, hi = -1;
+                    for (int b = seg; b < seg + 16; ++b)
+                        if (hist[f3 * 256 + b] & 0xFFFFu) {
+                            if (lo == 256) lo = b;
+                            hi = b;
+                        }
+                    if (hi >= 0) {
+                        atomicMin(&sh_bmin[f3], lo);
+                        atomicMax(&sh_bmax[f3], hi);
+                    }
+                }
+            }
+            __syncthreads();
+
+            // permutation + candidate walk (parallel draws)
+            if (tid < FPAD)
+                sh_draws[tid] = philox_draw(
+                    TAG_FEATSEL | ((uint32_t)(depth & 0xFF) << 8),
+                    (uint32_t)gs, (uint32_t)ge, (uint32_t)tid,
+                    a.seed, key);
+            __syncthreads();
+            if (tid == 0) {
+                int perm[FPAD];
+                for (int f = 0; f < F; ++f) perm[f] = f;
+                for (int i = 0; i < F - 1; ++i) {
+                    int j = i + (int)philox_bounded(sh_draws[i],
+                                                    (uint32_t)(F - i));
+                    int t = perm[i]; perm[i] = perm[j]; perm[j] = t;
+                }
+                int nc = 0;
+                for (int i = 0; i < F && nc < a.max_features; ++i) {
+                    int f = perm[i];
+                    if (sh_bmin[f] != sh_bmax[f]) sh_cand[nc++] = f;
+                }
+                sh_ncand = nc;
+            }
+            __syncthreads();
+
+            const int ncand = sh_ncand;
+            if (ncand == 0) {
+                __syncthreads();
+                continue;
+            }
+
+            // evaluate candidates (wave per candidate; identical math and
+            // tie-breaks to hist_split_kernel phase 5)
+            for (int ci = wave; ci < ncand; ci += HBLK / 64) {
+                const int f = sh_cand[ci];
+                const int bmin = sh_bmin[f], bmax = sh_bmax[f];
+                int ln[4], l1[4];
+                int tn = 0, t1 = 0;
+                for (int k = 0; k < 4; ++k) {
+                    uint32_t v = hist[f * 256 + lane * 4 + k];
+                    ln[k] = (int)(v & 0xFFFFu);
+                    l1[k] = (int)(v >> 16);
+                    tn += ln[k];
+                    t1 += l1[k];
+                }
+                int scn = tn, sc1 = t1;
+                for (int d = 1; d < 64; d <<= 1) {
+                    int un = __shfl_up(scn, d);
+                    int u1 = __shfl_up(sc1, d);
+                    if (lane >= d) { scn += un; sc1 += u1; }
+                }
+                const int excl_n = scn - tn, excl_1 = sc1 - t1;
+
+                double best_s = -1.0;
+                int best_b = -1, best_nl = 0;
+
+                if (a.splitter_random) {
+                    uint32_t tag = TAG_THRESH |
+                                   ((uint32_t)(depth & 0xFF) << 8);
+                    uint32_t u = philox_draw(tag, (uint32_t)gs,
+                                             (uint32_t)ge, (uint32_t)f,
+                                             a.seed, key);
+                    int b = bmin + (int)philox_bounded(
+                        u, (uint32_t)(bmax - bmin));
+                    if (b >= lane * 4 && b < lane * 4 + 4) {
+                        int cn = excl_n, c1f = excl_1;
+                        for (int k = 0; k <= b - lane * 4; ++k) {
+                            cn += ln[k];
+                            c1f += l1[k];
+                        }
+                        long nL = cn, n1L = c1f;
+                        long n0L = nL - n1L, nR = n - nL;
+                        long n1R = c1 - n1L, n0R = c0 - n0L;
+                        best_s = (double)(n0L * n0L + n1L * n1L)
+                                     / (double)nL
+                               + (double)(n0R * n0R + n1R * n1R)
+                                     / (double)nR;
+                        best_b = b;
+                        best_nl = (int)nL;
+                    }
+                } else {
+                    int cn = excl_n, c1f = excl_1;
+                    for (int k = 0; k < 4; ++k) {
+                        int b = lane * 4 + k;
+                        cn += ln[k];
+                        c1f += l1[k];
+                        if (ln[k] == 0 || b < bmin || b >= bmax) continue;
+                        long nL = cn, n1L = c1f;
+                        long n0L = nL - n1L, nR = n - nL;
+                        long n1R = c1 - n1L, n0R = c0 - n0L;
+                        double sc2 = (double)(n0L * n0L + n1L * n1L)
+                                         / (double)nL
+                                   + (double)(n0R * n0R + n1R * n1R)
+                                         / (double)nR;
+                        if (sc2 > best_s) {
+                            best_s = sc2; best_b = b; best_nl = (int)nL;
+                        }
+                    }
+                }
+                for (int d = 32; d > 0; d >>= 1) {
+                    double os = __shfl_down(best_s, d);
+                    int ob = __shfl_down(best_b, d);
+                    int onl = __shfl_down(best_nl, d);
+                    if (os > best_s || (os == best_s && ob != -1 &&
+                                        (best_b == -1 || ob < best_b))) {
+                        best_s = os; best_b = ob; best_nl = onl;
+                    }
+                }
+                if (lane == 0) {
+                    sh_score[ci] = best_s;
+                    sh_bin[ci] = best_b;
+                    sh_nL[ci] = best_nl;
+                }
+            }
+            __syncthreads();
+
+            if (tid == 0) {
+                double best_s = -1.0e300;
+                int bf = -1, bb = -1, bnl = 0;
+                for (int ci = 0; ci < ncand; ++ci) {
+                    if (sh_bin[ci] >= 0 && sh_score[ci] > best_s) {
+                        best_s = sh_score[ci];
+                        bf = sh_cand[ci];
+                        bb = sh_bin[ci];
+                        bnl = sh_nL[ci];
+                    }
+                }
+                sh_bestf = bf;
+                sh_bestbin = bb;
+                sh_bestnL = bnl;
+                if (bf >= 0) {
+                    int l = atomicAdd(&a.node_alloc[it.job], 2);
+                    a.nfeat[nbase + node] = bf;
+                    a.nsplit[nbase + node] = bb;
+                    a.nleft[nbase + node] = l;
+                    sh_lid = l;
+                }
+            }
+            __syncthreads();
+
+            const int bf = sh_bestf;
+            if (bf < 0) {
+                __syncthreads();
+                continue;
+            }
+            const int bb = sh_bestbin;
+            const int nL = sh_bestnL;
+
+            // stable partition of m_idx[ls,le) via m_idx2 (LDS)
+            if (tid == 0) {
+                sh_scan[HBLK - 1] = 0;   // reuse as running offsets
+            }
+            __shared__ int sh_lo, sh_ro;
+            if (tid == 0) { sh_lo = 0; sh_ro = 0; }
+            __syncthreads();
+            for (int base = ls; base < le; base += HBLK) {
+                const int i = base + tid;
+                const bool valid = i < le;
+                int o = 0, flag = 0;
+                if (valid) {
+                    o = m_idx[i];
+                    const uint4 cw = m_codes[o];
+                    const uint32_t wsel = (&cw.x)[bf >> 2];
+                    flag = (int)(((wsel >> ((bf & 3) * 8)) & 0xFFu)
+                                 <= (uint32_t)bb);
+                }
+                const unsigned long long lm = __ballot(valid && flag);
+                const unsigned long long below = (1ULL << lane) - 1ULL;
+                const int left_rank = __popcll(lm & below);
+                if (lane == 0) sh_scan[wave] = __popcll(lm);
+                __syncthreads();
+                int wave_left_excl = 0;
+                for (int ww = 0; ww < wave; ++ww)
+                    wave_left_excl += sh_scan[ww];
+                const int tile_left = sh_scan[0] + sh_scan[1] + sh_scan[2]
+                                      + sh_scan[3];
+                const int tile_n = min(HBLK, le - base);
+                if (valid) {
+                    const int dst = flag
+                        ? ls + sh_lo + wave_left_excl + left_rank
+                        : ls + nL + sh_ro + (i - base)
+                          - (wave_left_excl + left_rank);
+                    m_idx2[dst] = (uint16_t)o;
+                }
+                __syncthreads();
+                if (tid == 0) {
+                    sh_lo += tile_left;
+                    sh_ro += tile_n - tile_left;
+                }
+                __syncthreads();
+            }
+            for (int i = ls + tid; i < le; i += HBLK) m_idx[i] = m_idx2[i];
+            __syncthreads();
+
+            // route children: > SMALL_N continue in-block (smaller child
+            // processed first: push larger first); <= SMALL_N go to the
+            // small queue with their final global range
+            if (tid == 0) {
+                const int nR = n - nL;
+                const MidFrame left = {(short)ls, (short)(ls + nL),
+                                       depth + 1, sh_lid};
+                const MidFrame right = {(short)(ls + nL), (short)le,
+                                        depth + 1, sh_lid + 1};
+                const bool left_small = nL <= nR;
+                const MidFrame* ordered[2] = {
+                    left_small ? &right : &left,   // larger pushed first
+                    left_small ? &left : &right,
+                };
+                for (int x = 0; x < 2; ++x) {
+                    const MidFrame& fr = *ordered[x];
+                    const int fn = fr.e - fr.s;
+                    if (fn > SMALL_N) {
+                        stack[++sh_sp] = fr;
+                    } else {
+                        const int i = atomicAdd(a.small_count, 1);
+                        if (i < a.small_cap)
+                            a.small[i] = {it.job, fr.node,
+                                          it.start + fr.s, it.start + fr.e,
+                                          fr.depth, -1};
+                        else
+                            atomicExch(a.err_flag, 1);
+                    }
+                }
+            }
+            __syncthreads();
+        }
+
+        // write the final arrangement back so the small kernel (launched
+        // after this one) reads each farmed subtree's samples
+        __syncthreads();
+        for (int i = tid; i < n0; i += HBLK)
+            a.sidx_nxt[sbase + it.start + i] = m_rows[m_idx[i]];
         __syncthreads();
     }
 }

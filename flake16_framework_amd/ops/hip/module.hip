@@ -87,6 +87,12 @@ std::vector<at::Tensor> forest_fit(
                              codes.options().dtype(at::kByte));
     auto small_count = at::zeros({1}, opts_i32);
 
+    // mid-subtree queue: disjoint ranges of > SMALL_N samples each
+    const long mid_cap = S / (SMALL_N + 1) + 8;
+    auto mid_q = at::empty({mid_cap * (long)sizeof(WorkItem)},
+                           codes.options().dtype(at::kByte));
+    auto mid_count = at::zeros({1}, opts_i32);
+
     // Histogram-subtraction pools (see forest.hip): sized for the worst
     // per-level allocation, 2 slots per splitting node >= HIST_SAVE_MIN.
     int HIST_SAVE_MIN = 2048;
@@ -143,6 +149,9 @@ std::vector<at::Tensor> forest_fit(
     a.small = (WorkItem*)small_q.data_ptr();
     a.small_count = small_count.data_ptr<int>();
     a.small_cap = (int)(S + 2);
+    a.mid = (WorkItem*)mid_q.data_ptr();
+    a.mid_count = mid_count.data_ptr<int>();
+    a.mid_cap = (int)mid_cap;
 
     int GRID = 4096;
     if (const char* e = getenv("FLAKE16_FIT_GRID")) GRID = atoi(e);
@@ -166,6 +175,8 @@ std::vector<at::Tensor> forest_fit(
             a.nxt_count = counts.data_ptr<int>() + nx;
             CHECK_HIP(hipMemsetAsync(small_count.data_ptr<int>(), 0, 4,
                                      stream));
+            CHECK_HIP(hipMemsetAsync(mid_count.data_ptr<int>(), 0, 4,
+                                     stream));
             if (splitter_random)
                 et_split_kernel<<<GRID, HBLK, 0, stream>>>(a);
             else if (max_features < F && getenv("FLAKE16_RF_CANDONLY"))
@@ -174,6 +185,7 @@ std::vector<at::Tensor> forest_fit(
                 rf_cand_split_kernel<<<GRID, HBLK, 0, stream>>>(a);
             else
                 hist_split_kernel<<<GRID, HBLK, 0, stream>>>(a);
+            mid_subtree_kernel<<<2048, HBLK, 0, stream>>>(a, a.sidx_nxt);
             small_subtree_kernel<<<2048, HBLK, 0, stream>>>(
                 a, a.sidx_nxt);
             CHECK_HIP(hipMemcpyAsync(pinned_p + (c % PINSZ),
