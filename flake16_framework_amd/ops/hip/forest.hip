@@ -83,7 +83,7 @@ struct ForestDev {
 };
 
 #define SMALL_N 64
-#define MID_N 1024
+#define MID_N 2048
 
 // Three-way child routing: <=SMALL_N -> wave-subtree queue, <=MID_N ->
 // LDS-staged mid-subtree queue, else next level.  Called by one thread.
