@@ -1099,6 +1099,7 @@ __global__ void mid_subtree_kernel(ForestDev a,
     __shared__ int sh_bestf, sh_bestbin, sh_bestnL, sh_lid;
     __shared__ MidFrame stack[MID_STACK];
     __shared__ int sh_sp;
+    __shared__ int sh_lo, sh_ro;
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -1351,10 +1352,6 @@ __global__ void mid_subtree_kernel(ForestDev a,
             const int nL = sh_bestnL;
 
             // stable partition of m_idx[ls,le) via m_idx2 (LDS)
-            if (tid == 0) {
-                sh_scan[HBLK - 1] = 0;   // reuse as running offsets
-            }
-            __shared__ int sh_lo, sh_ro;
             if (tid == 0) { sh_lo = 0; sh_ro = 0; }
             __syncthreads();
             for (int base = ls; base < le; base += HBLK) {
