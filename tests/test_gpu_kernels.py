@@ -349,3 +349,24 @@ class TestTreeShapGpu:
         assert got.shape == ref.shape
         # scaling runs on device (fp64 reduction order differs) -> tolerance
         np.testing.assert_allclose(got, ref, atol=5e-4)
+
+
+class TestStreamDeterminism:
+    def test_stream_count_does_not_change_results(self, ops, monkeypatch):
+        """Per-cell results are bitwise identical whether cells run on one
+        stream or four (randomness is keyed on data identities, never on
+        scheduling)."""
+        from flake16_framework_amd.dataset.synthetic import (
+            make_synthetic_tests,
+        )
+        from flake16_framework_amd.engine.scores import run_scores
+
+        tests = make_synthetic_tests(n_tests=400, seed=2)
+        cells = [0, 1, 107]
+        monkeypatch.setenv("FLAKE16_STREAMS", "1")
+        r1 = run_scores(tests=tests, backend="hip", cells=cells)
+        monkeypatch.setenv("FLAKE16_STREAMS", "4")
+        r4 = run_scores(tests=tests, backend="hip", cells=cells)
+        for k in r1:
+            assert r1[k][2] == r4[k][2]
+            assert r1[k][3] == r4[k][3]
