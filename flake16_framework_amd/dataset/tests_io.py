@@ -58,12 +58,14 @@ def write_tests(tests_file=TESTS_FILE):
         json.dump(tests, fd, indent=4)
 
 
-def load_tests(tests_file=TESTS_FILE):
-    with open(tests_file, "r") as fd:
+def load_tests(tests_file=None):
+    # None means "the stage artifact in the working directory" — the CLI
+    # path passes no explicit file (reference behavior)
+    with open(tests_file or TESTS_FILE, "r") as fd:
         return json.load(fd)
 
 
-def load_feat_lab_proj(flaky_label, feature_set, tests_file=TESTS_FILE,
+def load_feat_lab_proj(flaky_label, feature_set, tests_file=None,
                        tests=None):
     """tests.json -> (features [N x len(feature_set)] float64,
     labels bool[N] (label == flaky_label), projects str[N]).
