@@ -141,14 +141,18 @@ __global__ void forest_init_kernel(
 // ---------------------------------------------------------------------------
 // The level kernel: histogram + split + partition, one workgroup per item.
 // ---------------------------------------------------------------------------
-// Histogram packing: one uint32 per (feature, bin): total count in bits
-// 0-15, class-1 count in bits 16-31.  One LDS atomic per
-// (sample, feature) instead of two, and 16 KiB LDS instead of 32 KiB
-// (double the resident blocks per CU).  Requires per-job n < 65536
-// (checked by the host driver).
+// Histogram packing (WIDE = false): one uint32 per (feature, bin) with
+// the total count in bits 0-15 and the class-1 count in bits 16-31 — one
+// LDS atomic per (sample, feature) and 16 KiB LDS (double the resident
+// blocks per CU).  Nodes with n >= 2^16 overflow the packing; the WIDE
+// instantiation uses two unpacked count planes (32 KiB LDS, two atomics)
+// and handles exactly those nodes — the two instantiations filter the
+// same work queue by node size, so arbitrarily large training sets work
+// (top levels wide, everything below packed).
+template <bool WIDE>
 __launch_bounds__(HBLK)
 __global__ void hist_split_kernel(ForestDev a) {
-    __shared__ uint32_t hist[FPAD * 256];
+    __shared__ uint32_t hist[FPAD * 256 * (WIDE ? 2 : 1)];
     __shared__ int sh_scan[HBLK];
     __shared__ int sh_bmin[FPAD], sh_bmax[FPAD];
     __shared__ int sh_cand[FPAD], sh_ncand;
@@ -169,14 +173,24 @@ __global__ void hist_split_kernel(ForestDev a) {
     for (int wi = blockIdx.x; wi < n_items; wi += gridDim.x) {
         WorkItem it = a.cur[wi];
         const int n = it.end - it.start;
+        if ((n >= 65536) != WIDE) continue;   // size-class filter (uniform)
         const long sbase = a.j_sidx_off[it.job];
         const long nbase = a.j_node_off[it.job];
         const uint32_t key = (uint32_t)a.j_key[it.job];
         const int F = a.F;
 
+        // helpers: per-(feature, bin) total / class-1 counts
+        auto h_n = [&](int f, int b) -> uint32_t {
+            return WIDE ? hist[f * 256 + b] : (hist[f * 256 + b] & 0xFFFFu);
+        };
+        auto h_1 = [&](int f, int b) -> uint32_t {
+            return WIDE ? hist[FPAD * 256 + f * 256 + b]
+                        : (hist[f * 256 + b] >> 16);
+        };
+
         // Phase 0/1: obtain the node histogram — either load the slot the
         // parent precomputed (subtraction scheme), or zero + accumulate.
-        if (it.hist_slot >= 0) {
+        if (!WIDE && it.hist_slot >= 0) {
             const uint32_t* src =
                 ((it.depth & 1) ? a.hist_pool1 : a.hist_pool0)
                 + (size_t)it.hist_slot * (FPAD * 256);
@@ -184,20 +198,23 @@ __global__ void hist_split_kernel(ForestDev a) {
                 reinterpret_cast<uint4*>(hist)[i] =
                     reinterpret_cast<const uint4*>(src)[i];
         } else {
-            for (int i = tid; i < F * 64; i += HBLK)
+            const int zwords = F * 64 * (WIDE ? 2 : 1);
+            for (int i = tid; i < zwords; i += HBLK)
                 reinterpret_cast<uint4*>(hist)[i] = uint4{0, 0, 0, 0};
             __syncthreads();
-            // One uint4 = the sample's 16 packed bin codes; one packed
-            // atomic per (sample, feature).
+            // One uint4 = the sample's 16 packed bin codes.
             for (int i = it.start + tid; i < it.end; i += HBLK) {
                 int row = a.sidx_cur[sbase + i];
                 uint4 cw = *reinterpret_cast<const uint4*>(
                     a.codes + (size_t)row * FPAD);
                 uint32_t w[4] = {cw.x, cw.y, cw.z, cw.w};
-                const uint32_t inc = 1u | ((uint32_t)a.labels[row] << 16);
+                const uint32_t lab = a.labels[row];
+                const uint32_t inc = WIDE ? 1u : (1u | (lab << 16));
                 for (int f = 0; f < F; ++f) {
                     uint32_t b = (w[f >> 2] >> ((f & 3) * 8)) & 0xFFu;
                     atomicAdd(&hist[f * 256 + b], inc);
+                    if (WIDE && lab)
+                        atomicAdd(&hist[FPAD * 256 + f * 256 + b], 1u);
                 }
             }
         }
@@ -206,7 +223,7 @@ __global__ void hist_split_kernel(ForestDev a) {
         // Phase 2: class counts from feature-0 histogram (wave reduce +
         // one cross-wave combine: 1 barrier instead of 8).
         {
-            int v = (int)(hist[tid] >> 16);
+            int v = (int)h_1(0, tid);
             for (int d = 32; d > 0; d >>= 1) v += __shfl_down(v, d);
             if (lane == 0) sh_scan[wave] = v;
         }
@@ -238,7 +255,7 @@ __global__ void hist_split_kernel(ForestDev a) {
             if (f3 < F) {
                 int lo = 256, hi = -1;
                 for (int b = seg; b < seg + 16; ++b)
-                    if (hist[f3 * 256 + b] & 0xFFFFu) {
+                    if (h_n(f3, b)) {
                         if (lo == 256) lo = b;
                         hi = b;
                     }
@@ -295,9 +312,8 @@ __global__ void hist_split_kernel(ForestDev a) {
             int tn = 0, t1 = 0;
             for (int k = 0; k < 4; ++k) {
                 int b = lane * 4 + k;
-                uint32_t v = hist[f * 256 + b];
-                ln[k] = (int)(v & 0xFFFFu);
-                l1[k] = (int)(v >> 16);
+                ln[k] = (int)h_n(f, b);
+                l1[k] = (int)h_1(f, b);
                 tn += ln[k];
                 t1 += l1[k];
             }
@@ -459,7 +475,7 @@ __global__ void hist_split_kernel(ForestDev a) {
             // class-1 count of the left child (prefix over chosen feature)
             int n1L = 0;
             for (int b = 0; b <= bb; ++b)
-                n1L += (int)(hist[bf * 256 + b] >> 16);
+                n1L += (int)h_1(bf, b);
             const int nR = n - nL;
             const int n1R = c1 - n1L;
             // only children that will run the histogram path need a slot
@@ -471,7 +487,10 @@ __global__ void hist_split_kernel(ForestDev a) {
 
             sh_slot_small = sh_slot_large = -1;
             sh_accum_small = 0;
-            if (n >= a.hist_save_min && (small_needs || large_needs)) {
+            // no pools on the WIDE path: packed 16-bit slots cannot hold
+            // its counts (its children > 2048 re-accumulate)
+            if (!WIDE && n >= a.hist_save_min &&
+                (small_needs || large_needs)) {
                 const int wp = (it.depth + 1) & 1;
                 const int want = (small_needs ? 1 : 0) +
                                  (large_needs ? 1 : 0);

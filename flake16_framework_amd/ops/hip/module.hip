@@ -50,10 +50,9 @@ std::vector<at::Tensor> forest_fit(
     const int* jn = j_n_cpu.data_ptr<int>();
     std::vector<long> sidx_off(J), node_off(J);
     long S = 0, Ntot = 0;
+    bool has_wide = false;   // any node >= 2^16 samples needs the WIDE path
     for (int j = 0; j < J; ++j) {
-        TORCH_CHECK(jn[j] < 65536,
-                    "forest_fit: per-job sample count must be < 65536 "
-                    "(packed 16-bit histogram counts)");
+        has_wide |= jn[j] >= 65536;
         sidx_off[j] = S;
         node_off[j] = Ntot;
         S += jn[j];
@@ -177,14 +176,18 @@ std::vector<at::Tensor> forest_fit(
                                      stream));
             CHECK_HIP(hipMemsetAsync(mid_count.data_ptr<int>(), 0, 4,
                                      stream));
-            if (splitter_random)
+            if (splitter_random) {
                 et_split_kernel<<<GRID, HBLK, 0, stream>>>(a);
-            else if (max_features < F && getenv("FLAKE16_RF_CANDONLY"))
+            } else if (!has_wide && max_features < F &&
+                       getenv("FLAKE16_RF_CANDONLY")) {
                 // ablation variant: candidate-only RF histograms measured
                 // ~3% SLOWER than full histograms + subtraction pools
                 rf_cand_split_kernel<<<GRID, HBLK, 0, stream>>>(a);
-            else
-                hist_split_kernel<<<GRID, HBLK, 0, stream>>>(a);
+            } else {
+                hist_split_kernel<false><<<GRID, HBLK, 0, stream>>>(a);
+                if (has_wide)
+                    hist_split_kernel<true><<<GRID, HBLK, 0, stream>>>(a);
+            }
             mid_subtree_kernel<<<2048, HBLK, 0, stream>>>(a, a.sidx_nxt);
             small_subtree_kernel<<<2048, HBLK, 0, stream>>>(
                 a, a.sidx_nxt);

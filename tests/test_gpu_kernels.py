@@ -370,3 +370,32 @@ class TestStreamDeterminism:
         for k in r1:
             assert r1[k][2] == r4[k][2]
             assert r1[k][3] == r4[k][3]
+
+
+class TestWideForest:
+    def test_ge_64k_samples_bitwise(self, ops, dev):
+        """Nodes >= 2^16 samples take the WIDE (unpacked) histogram path;
+        trees must still match the numpy reference bitwise."""
+        from flake16_framework_amd.models.binning import (
+            bin_codes, compute_bin_cuts,
+        )
+        from flake16_framework_amd.models.forest_ref import (
+            ForestParams, fit_forest,
+        )
+        X, y = _data(70000, seed=11, sep=1.0)
+        cuts = compute_bin_cuts(X)
+        codes = bin_codes(X, cuts)
+        params = ForestParams(2, True, "best", "sqrt", 0)
+        ref = fit_forest(codes, y, params, job_base=5, cuts=cuts)
+
+        codes_d = torch.from_numpy(np.ascontiguousarray(codes)).to(dev)
+        y_d = torch.from_numpy(y).to(dev)
+        out = ops.forest_fit(
+            codes_d, y_d,
+            torch.zeros(2, dtype=torch.int32, device=dev),
+            torch.full((2,), len(y), dtype=torch.int32, device=dev),
+            torch.arange(5, 7, dtype=torch.int32, device=dev),
+            16, 4, True, False, 0)
+        node_alloc = out[6].cpu().numpy()
+        assert node_alloc[0] == ref.trees[0].n_nodes
+        assert node_alloc[1] == ref.trees[1].n_nodes
