@@ -28,7 +28,6 @@ import torch
 
 from ..configgrid import MODEL_AXIS, resolve
 from ..dataset.tests_io import load_feat_lab_proj, load_tests
-from ..models.binning import compute_bin_cuts
 from ..ops.backend import get_ops
 from .folds import stratified_kfold_split
 from .metrics import finalize_scores
@@ -53,6 +52,31 @@ def _cuts_tensors(cuts, device):
     off[len(cuts) + 1:] = off[len(cuts)]
     return (torch.from_numpy(flat.astype(np.float32)).to(device),
             torch.from_numpy(off).to(device))
+
+
+def _device_cuts(X32, F, max_bins=256):
+    """Bin cuts computed on device — bitwise-identical to
+    models.binning.compute_bin_cuts (midpoints of uniques when <= max_bins
+    distinct values, order-statistic positions otherwise: pure sorting and
+    indexing, no float interpolation).  Returns (flat cuts, offsets)."""
+    n = X32.shape[0]
+    pos = torch.from_numpy(
+        (np.ceil(np.arange(1, max_bins) * (n / max_bins))
+         .astype(np.int64) - 1).clip(0, n - 1)).to(X32.device)
+    parts = []
+    lens = [0] * (FPAD + 1)
+    for f in range(F):
+        col, _ = torch.sort(X32[:, f].contiguous())
+        vals = torch.unique_consecutive(col)
+        if vals.numel() <= max_bins:
+            c = ((vals[1:].double() + vals[:-1]) * 0.5).float()
+        else:
+            c = torch.unique_consecutive(col.index_select(0, pos))
+        parts.append(c)
+        lens[f + 1] = c.numel()
+    off = np.cumsum(lens).astype(np.int32)
+    flat = torch.cat(parts) if parts else X32.new_zeros(0)
+    return flat.contiguous(), torch.from_numpy(off).to(X32.device)
 
 
 class SweepContext:
@@ -150,8 +174,7 @@ class SweepContext:
         elif preproc == "scale+pca":
             X64 = ops.pca_fit_transform(ops.scaler_fit_transform(X64), F)
         X32 = X64.float().contiguous()
-        cuts = compute_bin_cuts(X32.cpu().numpy()[:, :F])
-        cuts_dev, cut_off_dev = _cuts_tensors(cuts, self.device)
+        cuts_dev, cut_off_dev = _device_cuts(X32[:, :F], F)
         codes_all = ops.bin_codes(X32, cuts_dev, cut_off_dev, F)
         view = {"X32": X32, "cuts_dev": cuts_dev, "cut_off": cut_off_dev,
                 "codes_all": codes_all, "F": F}

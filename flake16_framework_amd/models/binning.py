@@ -27,15 +27,21 @@ def compute_bin_cuts(X, max_bins=MAX_BINS):
     quantile cuts.
     """
     X = np.asarray(X)
+    n = X.shape[0]
     cuts = []
     for f in range(X.shape[1]):
         vals = np.unique(X[:, f].astype(np.float32))
         if len(vals) <= max_bins:
             c = ((vals[1:].astype(np.float64) + vals[:-1]) * 0.5).astype(np.float32)
         else:
-            qs = np.quantile(X[:, f].astype(np.float64),
-                             np.linspace(0.0, 1.0, max_bins + 1)[1:-1])
-            c = np.unique(qs.astype(np.float32))
+            # order-statistic cuts: pure indexing into the sorted column —
+            # no float interpolation, so the GPU path (torch.sort + the
+            # same positions, engine/hip_cell._device_cuts) produces
+            # BITWISE-identical cut arrays
+            xs = np.sort(X[:, f].astype(np.float32))
+            pos = (np.ceil(np.arange(1, max_bins) * (n / max_bins))
+                   .astype(np.int64) - 1).clip(0, n - 1)
+            c = np.unique(xs[pos])
         cuts.append(np.ascontiguousarray(c, dtype=np.float32))
     return cuts
 

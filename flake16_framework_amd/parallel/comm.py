@@ -65,17 +65,25 @@ def shard_cells(world, rank, n_cells=None):
         keys = keys[:n_cells]
 
     groups = {}
+    gview = {}
     for i, k in enumerate(keys):
-        groups.setdefault(balance_group_index(k), []).append(i)
+        g = balance_group_index(k)
+        groups.setdefault(g, []).append(i)
+        gview[g] = (k[1], k[2])   # (feature-set, preprocessing) view
     gcost = {g: sum(cell_cost_estimate(keys[i]) for i in cells)
              for g, cells in groups.items()}
-    order = sorted(groups, key=lambda g: (-gcost[g], g))
 
-    load = [0.0] * world
+    # VIEW-AFFINE sequential fill: groups ordered by their preprocessed
+    # view, ranks take contiguous cost slices — each rank touches only
+    # 1-2 of the 6 views, so the per-rank view-build fixed cost (which
+    # does not shrink with world size) stays small at 8 GPUs.
+    order = sorted(groups, key=lambda g: (gview[g], -gcost[g], g))
+    total = sum(gcost.values())
     mine = []
+    acc = 0.0
     for g in order:
-        r = min(range(world), key=lambda j: (load[j], j))
-        load[r] += gcost[g]
+        r = min(int(acc / total * world), world - 1) if total else 0
+        acc += gcost[g]
         if r == rank:
             mine.extend(groups[g])
 
