@@ -86,8 +86,26 @@ def compute_shap_hip(config_keys, idx, tests=None, tests_file=None, seed=0):
                        max_features, spec["bootstrap"],
                        spec["kind"] == "extra_trees", seed)
 
-    phi = ops.treeshap(view["codes_all"], j_node_off, nfeat, nsplit, nleft,
-                       ncnt0, ncnt1)
+    import os
+    if os.environ.get("FLAKE16_SHAP_RECURSE"):
+        # the original per-(sample, tree) recursion — kept for A/B
+        phi = ops.treeshap(view["codes_all"], j_node_off, nfeat, nsplit,
+                           nleft, ncnt0, ncnt1)
+    else:
+        # leaf-path formulation: wave-uniform walks, no path copies
+        from ..models.leafpaths import build_leaf_paths
+        leaf_tree, leaf_off, path_nodes, max_depth = build_leaf_paths(
+            nfeat.cpu().numpy(), nleft.cpu().numpy(),
+            j_node_off.cpu().numpy(), node_alloc.cpu().numpy())
+        if max_depth > 128:
+            raise RuntimeError(
+                f"treeshap: tree depth {max_depth} exceeds SHAP_DMAX")
+        phi = ops.treeshap_paths(
+            view["codes_all"],
+            torch.from_numpy(leaf_tree).to(device),
+            torch.from_numpy(leaf_off).to(device),
+            torch.from_numpy(path_nodes).to(device),
+            j_node_off, nfeat, nsplit, nleft, ncnt0, ncnt1)
     return (phi.cpu().numpy() / n_trees)[:, :F]
 
 

@@ -419,9 +419,32 @@ at::Tensor treeshap(at::Tensor codes, at::Tensor j_node_off,
     return phi;
 }
 
+at::Tensor treeshap_paths(at::Tensor codes, at::Tensor leaf_tree,
+                          at::Tensor leaf_off, at::Tensor path_nodes,
+                          at::Tensor j_node_off, at::Tensor nfeat,
+                          at::Tensor nsplit, at::Tensor nleft,
+                          at::Tensor ncnt0, at::Tensor ncnt1) {
+    const at::cuda::OptionalCUDAGuard guard(codes.device());
+    const int n_samples = codes.size(0);
+    const int n_leaves = leaf_tree.size(0);
+    auto phi = at::zeros({n_samples, 16},
+                         codes.options().dtype(at::kDouble));
+    treeshap_paths_kernel<<<2048, SHAP_BLK, 0, current_stream()>>>(
+        codes.data_ptr<uint8_t>(), n_samples, leaf_tree.data_ptr<int>(),
+        leaf_off.data_ptr<int>(), path_nodes.data_ptr<int>(),
+        j_node_off.data_ptr<long>(), nfeat.data_ptr<int>(),
+        nsplit.data_ptr<int>(), nleft.data_ptr<int>(),
+        ncnt0.data_ptr<float>(), ncnt1.data_ptr<float>(), n_leaves,
+        phi.data_ptr<double>());
+    return phi;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("treeshap", &treeshap, py::call_guard<py::gil_scoped_release>(),
           "Path-dependent TreeSHAP (class-0), summed over trees");
+    m.def("treeshap_paths", &treeshap_paths,
+          py::call_guard<py::gil_scoped_release>(),
+          "Leaf-path TreeSHAP (class-0), summed over trees");
     // gil_scoped_release: the forest_fit level loop blocks on stream syncs;
     // releasing the GIL lets other Python threads drive their own streams.
     m.def("forest_fit", &forest_fit,

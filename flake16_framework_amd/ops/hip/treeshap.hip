@@ -230,3 +230,94 @@ __global__ void tree_depth_kernel(const long* __restrict__ j_node_off,
     }
     depth_out[t] = dmax;
 }
+
+// ---------------------------------------------------------------------------
+// Leaf-path formulation: one thread per (leaf, sample) pair, pairs ordered
+// leaf-major so a wave's 64 lanes walk the SAME root->leaf path for 64
+// consecutive samples — control flow is wave-uniform and there are no
+// per-frame path copies or stacks.  The EXTEND/UNWIND sequence along the
+// path reproduces the recursion's math exactly (the recursion unwinds a
+// repeated feature before descending, which is step order along the path).
+// ---------------------------------------------------------------------------
+#define SHAP_DMAX 128
+
+__global__ void treeshap_paths_kernel(
+    const uint8_t* __restrict__ codes,       // [n_samples, 16]
+    int n_samples,
+    const int* __restrict__ leaf_tree,       // [L] tree of leaf
+    const int* __restrict__ leaf_off,        // [L+1] CSR into path_nodes
+    const int* __restrict__ path_nodes,      // concatenated root->leaf ids
+    const long* __restrict__ j_node_off,     // [n_trees]
+    const int* __restrict__ nfeat, const int* __restrict__ nsplit,
+    const int* __restrict__ nleft,
+    const float* __restrict__ ncnt0, const float* __restrict__ ncnt1,
+    int n_leaves,
+    double* __restrict__ phi /* [n_samples, 16] */) {
+    const long n_pairs = (long)n_leaves * n_samples;
+    const long stride = (long)gridDim.x * blockDim.x;
+
+    for (long pair = (long)blockIdx.x * blockDim.x + threadIdx.x;
+         pair < n_pairs; pair += stride) {
+        const int li = (int)(pair / n_samples);
+        const int si = (int)(pair % n_samples);
+        const uint8_t* cr = codes + (size_t)si * 16;
+        const long nb = j_node_off[leaf_tree[li]];
+        const int p0 = leaf_off[li];
+        const int plen = leaf_off[li + 1] - p0;   // nodes incl. the leaf
+        if (plen > SHAP_DMAX) continue;           // guarded by the host
+
+        PathElem m[SHAP_DMAX + 1];
+        int l = 0;
+        double pz = 1.0, po = 1.0;
+        int pi = -1;
+
+        for (int step = 0; step < plen; ++step) {
+            const int node = path_nodes[p0 + step];
+
+            // unwind a repeated feature BEFORE extending with this step's
+            // accumulated (pz, po) — matching the recursion, the repeat
+            // check and unwind happen at the PARENT, i.e. they went into
+            // the (pz, po) carried into this extend; so: extend first,
+            // then (for internal nodes) prepare the child's (pz, po).
+            l = shap_extend(m, l, pz, po, pi);
+
+            const int f = nfeat[nb + node];
+            if (f < 0) break;   // the leaf itself: extended, done
+
+            double iz = 1.0, io = 1.0;
+            int k = -1;
+            for (int i = 0; i < l; ++i)
+                if (m[i].d == f) { k = i; break; }
+            if (k >= 0) {
+                iz = m[k].z;
+                io = m[k].o;
+                l = shap_unwind(m, l, k);
+            }
+
+            const int child = path_nodes[p0 + step + 1];
+            const int lc = nleft[nb + node];
+            const int hot = lc + ((int)cr[f] <= nsplit[nb + node] ? 0 : 1);
+            const double rj = (double)ncnt0[nb + node] +
+                              (double)ncnt1[nb + node];
+            const double rv = (double)ncnt0[nb + child] +
+                              (double)ncnt1[nb + child];
+            pz = iz * rv / rj;
+            po = (child == hot) ? io : 0.0;
+            pi = f;
+        }
+
+        const int leaf = path_nodes[p0 + plen - 1];
+        const double c0 = (double)ncnt0[nb + leaf];
+        const double c1 = (double)ncnt1[nb + leaf];
+        const double v = c0 / (c0 + c1);
+
+        double lphi[17];
+        for (int f = 0; f < 17; ++f) lphi[f] = 0.0;
+        for (int i = 1; i < l; ++i)
+            lphi[m[i].d + 1] +=
+                shap_unwound_sum(m, l, i) * (m[i].o - m[i].z) * v;
+        for (int f = 0; f < 16; ++f)
+            if (lphi[f + 1] != 0.0)
+                atomicAdd(&phi[(size_t)si * 16 + f], lphi[f + 1]);
+    }
+}

@@ -399,3 +399,42 @@ class TestWideForest:
         node_alloc = out[6].cpu().numpy()
         assert node_alloc[0] == ref.trees[0].n_nodes
         assert node_alloc[1] == ref.trees[1].n_nodes
+
+
+class TestTreeShapPaths:
+    def test_leafpath_kernel_matches_ref(self, ops, dev):
+        from flake16_framework_amd.models.binning import (
+            bin_codes, compute_bin_cuts,
+        )
+        from flake16_framework_amd.models.forest_ref import (
+            ForestParams, fit_forest,
+        )
+        from flake16_framework_amd.models.leafpaths import build_leaf_paths
+        from flake16_framework_amd.models.treeshap_ref import forest_shap
+
+        X, y = _data(300, f=16, seed=9)
+        cuts = compute_bin_cuts(X)
+        codes = bin_codes(X, cuts)
+        params = ForestParams(10, True, "best", "sqrt", 0)
+        forest = fit_forest(codes, y, params, job_base=40, cuts=cuts)
+        ref = forest_shap(forest, codes[:50], 16)
+
+        codes_d = torch.from_numpy(np.ascontiguousarray(codes)).to(dev)
+        y_d = torch.from_numpy(y).to(dev)
+        nfeat, nsplit, nleft, ncnt0, ncnt1, j_node_off, node_alloc = \
+            ops.forest_fit(
+                codes_d, y_d,
+                torch.zeros(10, dtype=torch.int32, device=dev),
+                torch.full((10,), len(y), dtype=torch.int32, device=dev),
+                torch.arange(40, 50, dtype=torch.int32, device=dev),
+                16, 4, True, False, 0)
+        leaf_tree, leaf_off, path_nodes, _ = build_leaf_paths(
+            nfeat.cpu().numpy(), nleft.cpu().numpy(),
+            j_node_off.cpu().numpy(), node_alloc.cpu().numpy())
+        phi = ops.treeshap_paths(
+            codes_d[:50].contiguous(),
+            torch.from_numpy(leaf_tree).to(dev),
+            torch.from_numpy(leaf_off).to(dev),
+            torch.from_numpy(path_nodes).to(dev),
+            j_node_off, nfeat, nsplit, nleft, ncnt0, ncnt1)
+        np.testing.assert_allclose(phi.cpu().numpy() / 10, ref, atol=1e-9)

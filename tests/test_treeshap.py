@@ -66,3 +66,35 @@ class TestTreeShapRef:
         shap = forest_shap(forest, codes[:10], 6)
         assert shap.shape == (10, 6)
         assert np.isfinite(shap).all()
+
+
+class TestLeafPaths:
+    def test_csr_paths_valid(self):
+        import numpy as np
+        from flake16_framework_amd.models.leafpaths import build_leaf_paths
+        forest, codes = _fit_small(n=200, f=8, seed=4, n_trees=3,
+                                   bootstrap=True)
+        nfeat = np.concatenate([t.feature for t in forest.trees])
+        nleft = np.concatenate([t.left for t in forest.trees])
+        offs, alloc = [], []
+        base = 0
+        for t in forest.trees:
+            offs.append(base)
+            alloc.append(t.n_nodes)
+            base += t.n_nodes
+        leaf_tree, leaf_off, path_nodes, max_d = build_leaf_paths(
+            nfeat, nleft, np.array(offs), np.array(alloc))
+
+        n_leaves = sum(int((t.feature == -1).sum()) for t in forest.trees)
+        assert len(leaf_tree) == n_leaves
+        for li in range(len(leaf_tree)):
+            t = forest.trees[leaf_tree[li]]
+            path = path_nodes[leaf_off[li]:leaf_off[li + 1]]
+            assert path[0] == 0
+            for i in range(len(path) - 1):
+                u, v = path[i], path[i + 1]
+                assert t.feature[u] != -1
+                assert v in (t.left[u], t.left[u] + 1)
+            assert t.feature[path[-1]] == -1
+        assert max_d == max(len(path_nodes[leaf_off[i]:leaf_off[i + 1]])
+                            for i in range(len(leaf_tree)))
