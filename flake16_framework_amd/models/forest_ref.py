@@ -268,18 +268,29 @@ def _tree_leaves(tree, codes_test):
                               tree.right[node[rows]])
 
 
+PREDICT_TREE_BLOCK = 10
+
+
 def predict_forest(forest, codes_test):
-    """Ensemble prediction: mean per-tree class probabilities, fp64
-    accumulation in tree order, ties -> class 0 (sklearn argmax)."""
+    """Ensemble prediction: mean per-tree class probabilities, ties ->
+    class 0 (sklearn argmax).  fp64 accumulation is BLOCKED in groups of
+    PREDICT_TREE_BLOCK trees (sequential within a block, blocks summed in
+    ascending order) — the exact association the device predict kernel
+    uses, so predictions stay bit-identical across backends."""
     codes_test = np.asarray(codes_test, dtype=np.uint8)
     m = codes_test.shape[0]
     acc0 = np.zeros(m, dtype=np.float64)
     acc1 = np.zeros(m, dtype=np.float64)
 
-    for tree in forest.trees:
-        leaf = _tree_leaves(tree, codes_test)
-        tot = tree.count0[leaf] + tree.count1[leaf]
-        acc0 += tree.count0[leaf] / tot
-        acc1 += tree.count1[leaf] / tot
+    for b in range(0, len(forest.trees), PREDICT_TREE_BLOCK):
+        b0 = np.zeros(m, dtype=np.float64)
+        b1 = np.zeros(m, dtype=np.float64)
+        for tree in forest.trees[b:b + PREDICT_TREE_BLOCK]:
+            leaf = _tree_leaves(tree, codes_test)
+            tot = tree.count0[leaf] + tree.count1[leaf]
+            b0 += tree.count0[leaf] / tot
+            b1 += tree.count1[leaf] / tot
+        acc0 += b0
+        acc1 += b1
 
     return (acc1 > acc0).astype(np.uint8)

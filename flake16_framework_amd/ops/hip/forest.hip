@@ -1672,13 +1672,85 @@ __global__ void small_subtree_kernel(ForestDev a,
 }
 
 // ---------------------------------------------------------------------------
-// Ensemble prediction + confusion.
-//
-// One thread per (fold, test-row) pair; trees visited in job order so the
-// fp64 probability accumulation matches forest_ref.predict_forest exactly.
-// Confusion counts go to per-(project, k) atomics, k = 2*y + pred - 1 with
-// true negatives skipped (reference experiment.py:476-483).
+// Ensemble prediction + confusion, two stages for parallelism:
+//   stage 1: one thread per (pair, tree-block of PREDICT_TREE_BLOCK) —
+//            partial fp64 probability sums (sequential within the block);
+//   stage 2: one thread per pair — blocks summed in ASCENDING order
+//            (deterministic; forest_ref.predict_forest uses the same
+//            association), then the confusion atomics
+//            (k = 2*y + pred - 1, true negatives skipped; reference
+//            experiment.py:476-483).
 // ---------------------------------------------------------------------------
+#define PREDICT_TREE_BLOCK 10
+
+__global__ void predict_partial_kernel(
+    const uint8_t* __restrict__ codes_test,   // [M, FPAD]
+    const int* __restrict__ pair_row,         // [P]
+    const int* __restrict__ pair_fold,        // [P]
+    int n_pairs,
+    const long* __restrict__ j_node_off,      // [J]
+    const int* __restrict__ nfeat, const int* __restrict__ nsplit,
+    const int* __restrict__ nleft,
+    const float* __restrict__ ncnt0, const float* __restrict__ ncnt1,
+    int trees_per_fold, int n_blocks,
+    double* __restrict__ partial /* [P, n_blocks, 2] */) {
+    const long total = (long)n_pairs * n_blocks;
+    const long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= total) return;
+    const int p = (int)(idx % n_pairs);
+    const int tb = (int)(idx / n_pairs);
+
+    const int row = pair_row[p];
+    const int fold = pair_fold[p];
+    const uint8_t* cr = codes_test + (size_t)row * FPAD;
+
+    const int t0 = tb * PREDICT_TREE_BLOCK;
+    const int t1 = min(trees_per_fold, t0 + PREDICT_TREE_BLOCK);
+    double acc0 = 0.0, acc1 = 0.0;
+    for (int t = t0; t < t1; ++t) {
+        const long nbase = j_node_off[fold * trees_per_fold + t];
+        int node = 0;
+        int f = nfeat[nbase];
+        while (f != LEAF_SENTINEL) {
+            int go_right = (int)cr[f] > nsplit[nbase + node];
+            node = nleft[nbase + node] + go_right;
+            f = nfeat[nbase + node];
+        }
+        const double c0 = (double)ncnt0[nbase + node];
+        const double c1 = (double)ncnt1[nbase + node];
+        const double tot = c0 + c1;
+        acc0 += c0 / tot;
+        acc1 += c1 / tot;
+    }
+    partial[((size_t)p * n_blocks + tb) * 2 + 0] = acc0;
+    partial[((size_t)p * n_blocks + tb) * 2 + 1] = acc1;
+}
+
+__global__ void predict_combine_kernel(
+    const uint8_t* __restrict__ y_test,       // [M]
+    const int* __restrict__ proj_id,          // [M]
+    const int* __restrict__ pair_row, int n_pairs,
+    const double* __restrict__ partial, int n_blocks,
+    uint8_t* __restrict__ pred_out,
+    int* __restrict__ confusion, int n_proj) {
+    const int p = blockIdx.x * blockDim.x + threadIdx.x;
+    if (p >= n_pairs) return;
+    double acc0 = 0.0, acc1 = 0.0;
+    for (int tb = 0; tb < n_blocks; ++tb) {
+        acc0 += partial[((size_t)p * n_blocks + tb) * 2 + 0];
+        acc1 += partial[((size_t)p * n_blocks + tb) * 2 + 1];
+    }
+    const int pred = acc1 > acc0;
+    pred_out[p] = (uint8_t)pred;
+    const int row = pair_row[p];
+    const int k = 2 * (int)y_test[row] + pred - 1;
+    if (k >= 0) {
+        atomicAdd(&confusion[proj_id[row] * 3 + k], 1);
+        atomicAdd(&confusion[n_proj * 3 + k], 1);
+    }
+}
+
+// (single-stage original, no longer launched — kept for reference builds)
 __global__ void predict_confusion_kernel(
     const uint8_t* __restrict__ codes_test,   // [M, FPAD] full-dataset codes
     const uint8_t* __restrict__ y_test,       // [M]

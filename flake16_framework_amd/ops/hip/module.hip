@@ -219,17 +219,25 @@ std::vector<at::Tensor> forest_predict_confusion(
     int64_t n_proj) {
     const at::cuda::OptionalCUDAGuard guard(codes_test.device());
     const int P = pair_row.size(0);
+    const int n_blocks =
+        ((int)trees_per_fold + PREDICT_TREE_BLOCK - 1) / PREDICT_TREE_BLOCK;
     auto pred = at::zeros({P}, codes_test.options().dtype(at::kByte));
     auto confusion = at::zeros({n_proj + 1, 3},
                                codes_test.options().dtype(at::kInt));
-    const int grid = (P + 255) / 256;
-    predict_confusion_kernel<<<grid, 256, 0, current_stream()>>>(
-        codes_test.data_ptr<uint8_t>(), y_test.data_ptr<uint8_t>(),
-        proj_id.data_ptr<int>(), pair_row.data_ptr<int>(),
+    auto partial = at::empty({(long)P * n_blocks * 2},
+                             codes_test.options().dtype(at::kDouble));
+    hipStream_t s = current_stream();
+    const long total = (long)P * n_blocks;
+    predict_partial_kernel<<<(int)((total + 255) / 256), 256, 0, s>>>(
+        codes_test.data_ptr<uint8_t>(), pair_row.data_ptr<int>(),
         pair_fold.data_ptr<int>(), P, j_node_off.data_ptr<long>(),
         nfeat.data_ptr<int>(), nsplit.data_ptr<int>(),
         nleft.data_ptr<int>(), ncnt0.data_ptr<float>(),
-        ncnt1.data_ptr<float>(), (int)trees_per_fold,
+        ncnt1.data_ptr<float>(), (int)trees_per_fold, n_blocks,
+        partial.data_ptr<double>());
+    predict_combine_kernel<<<(P + 255) / 256, 256, 0, s>>>(
+        y_test.data_ptr<uint8_t>(), proj_id.data_ptr<int>(),
+        pair_row.data_ptr<int>(), P, partial.data_ptr<double>(), n_blocks,
         pred.data_ptr<uint8_t>(), confusion.data_ptr<int>(), (int)n_proj);
     return {pred, confusion};
 }
