@@ -45,15 +45,6 @@ def _pad16(X):
     return out.contiguous()
 
 
-def _cuts_tensors(cuts, device):
-    flat = np.concatenate(cuts) if len(cuts) else np.zeros(0, np.float32)
-    off = np.zeros(FPAD + 1, dtype=np.int32)
-    off[1:len(cuts) + 1] = np.cumsum([len(c) for c in cuts])
-    off[len(cuts) + 1:] = off[len(cuts)]
-    return (torch.from_numpy(flat.astype(np.float32)).to(device),
-            torch.from_numpy(off).to(device))
-
-
 def _device_cuts(X32, F, max_bins=256):
     """Bin cuts computed on device — bitwise-identical to
     models.binning.compute_bin_cuts (midpoints of uniques when <= max_bins

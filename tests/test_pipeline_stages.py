@@ -282,3 +282,54 @@ class TestCli:
             main(["bogus"])
         with pytest.raises(ValueError):
             main([])
+
+
+class TestReportHelpers:
+    def test_cell_formatters(self):
+        import numpy as np
+        from flake16_framework_amd.report.figures import (
+            cellfn_corr, cellfn_default, cellfn_shap,
+        )
+        assert cellfn_default("x") == "x"
+        assert cellfn_default(0.125) == "0.12"
+        assert cellfn_default(0) == "-"
+        assert cellfn_default(np.int64(7)) == "7"
+        assert cellfn_corr(-0.5) == "\\cellcolor{gray!25} -0.50"
+        assert cellfn_shap(0.12345) == "0.123"
+
+    def test_write_table_layout(self, tmp_path):
+        from flake16_framework_amd.report.figures import write_table
+        p = tmp_path / "t.tex"
+        write_table(str(p), [[["a", 1.0], ["b", 2.0]], [["c", 3.0]]])
+        text = p.read_text()
+        assert "\\midrule" in text            # between blocks
+        assert "\\rowcolor{gray!20}" in text  # zebra on odd rows
+        assert "a & 1.00 \\\\" in text
+
+    def test_req_runs_plot_coords_normalized(self):
+        from flake16_framework_amd.report.figures import (
+            get_req_runs_plot_coords,
+        )
+        coords = get_req_runs_plot_coords({50: 2, 150: 1, 2400: 1})
+        pairs = [c.strip("()").split(",") for c in coords.split(" ")]
+        assert len(pairs) == 25
+        assert float(pairs[-1][1]) == 1.0     # normalized by the final bin
+        assert float(pairs[0][1]) == 0.5      # 2 of 4 within 100 runs
+
+
+class TestChurnCollector:
+    def test_git_log_hunks_counted(self, tmp_path):
+        import subprocess
+        from flake16_framework_amd.collect.testinspect import collect_churn
+        git = ["git", "-C", str(tmp_path), "-c", "user.email=t@t",
+               "-c", "user.name=t"]
+        subprocess.run([*git[:3], "init", "-q"], check=True)
+        f = tmp_path / "a.py"
+        f.write_text("one\ntwo\nthree\n")
+        subprocess.run([*git, "add", "a.py"], check=True)
+        subprocess.run([*git, "commit", "-q", "-m", "c1"], check=True)
+        f.write_text("one\nTWO\nthree\n")
+        subprocess.run([*git, "commit", "-q", "-am", "c2"], check=True)
+        churn = collect_churn(str(tmp_path), None)
+        assert churn["a.py"][2] >= 2   # line 2 touched by both commits
+        assert churn["a.py"][1] >= 1
