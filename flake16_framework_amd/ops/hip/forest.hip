@@ -65,7 +65,7 @@ struct ForestDev {
     // stored in pool[(depth+1)&1] and the children skip accumulation.
     uint32_t* __restrict__ hist_pool0;
     uint32_t* __restrict__ hist_pool1;
-    int* __restrict__ pool_count;   // [2], write-parity slot zeroed per level
+    int* __restrict__ pool_count;   // strided level-state slots (see host)
     int pool_cap;
     int hist_save_min;
     // Small-subtree routing: children with n <= SMALL_N go to this queue
@@ -494,7 +494,7 @@ __global__ void hist_split_kernel(ForestDev a) {
                 const int wp = (it.depth + 1) & 1;
                 const int want = (small_needs ? 1 : 0) +
                                  (large_needs ? 1 : 0);
-                int s = atomicAdd(&a.pool_count[wp], want);
+                int s = atomicAdd(&a.pool_count[wp * 4], want);
                 if (s + want <= a.pool_cap) {
                     sh_accum_small = 1;
                     if (small_needs) sh_slot_small = s++;

@@ -66,10 +66,12 @@ std::vector<at::Tensor> forest_fit(
                           .to(codes.device());
 
     auto nfeat = at::full({Ntot}, LEAF_SENTINEL, opts_i32);
-    auto nsplit = at::zeros({Ntot}, opts_i32);
-    auto nleft = at::zeros({Ntot}, opts_i32);
-    auto ncnt0 = at::zeros({Ntot}, opts_f32);
-    auto ncnt1 = at::zeros({Ntot}, opts_f32);
+    // only nfeat needs initialization (LEAF sentinel); the other node
+    // fields are written for every reached node before any read
+    auto nsplit = at::empty({Ntot}, opts_i32);
+    auto nleft = at::empty({Ntot}, opts_i32);
+    auto ncnt0 = at::empty({Ntot}, opts_f32);
+    auto ncnt1 = at::empty({Ntot}, opts_f32);
     auto node_alloc = at::zeros({J}, opts_i32);
 
     auto sidx_a = at::empty({S}, opts_i32);
@@ -79,18 +81,19 @@ std::vector<at::Tensor> forest_fit(
                             codes.options().dtype(at::kByte));
     auto work_b = at::empty({work_cap * (long)sizeof(WorkItem)},
                             codes.options().dtype(at::kByte));
-    auto counts = at::zeros({2}, opts_i32);
+    // per-level counter state, one 16 B memset per level:
+    // state[parity*4 + 0] = work count, +1 = pool count, +2 = small,
+    // +3 = mid
+    auto state = at::zeros({8}, opts_i32);
     auto err = at::zeros({1}, opts_i32);
 
     auto small_q = at::empty({(S + 2) * (long)sizeof(WorkItem)},
                              codes.options().dtype(at::kByte));
-    auto small_count = at::zeros({1}, opts_i32);
 
     // mid-subtree queue: disjoint ranges of > SMALL_N samples each
     const long mid_cap = S / (SMALL_N + 1) + 8;
     auto mid_q = at::empty({mid_cap * (long)sizeof(WorkItem)},
                            codes.options().dtype(at::kByte));
-    auto mid_count = at::zeros({1}, opts_i32);
 
     // Histogram-subtraction pools (see forest.hip): sized for the worst
     // per-level allocation, 2 slots per splitting node >= HIST_SAVE_MIN.
@@ -103,7 +106,6 @@ std::vector<at::Tensor> forest_fit(
                                 codes.options().dtype(at::kInt));
     auto hist_pool1 = at::empty({pool_cap * FPAD * 256},
                                 codes.options().dtype(at::kInt));
-    auto pool_count = at::zeros({2}, opts_i32);
 
     hipStream_t stream = current_stream();
 
@@ -112,7 +114,7 @@ std::vector<at::Tensor> forest_fit(
         j_sidx_off.data_ptr<long>(), j_key.data_ptr<int>(),
         node_alloc.data_ptr<int>(), sidx_a.data_ptr<int>(),
         (WorkItem*)work_a.data_ptr(), bootstrap ? 1 : 0, (uint32_t)seed);
-    counts.narrow(0, 0, 1).fill_((int)J);
+    state.narrow(0, 0, 1).fill_((int)J);
 
     const int PINSZ = 64;
     auto pinned = at::empty({PINSZ}, at::TensorOptions()
@@ -142,15 +144,14 @@ std::vector<at::Tensor> forest_fit(
     a.work_cap = (int)work_cap;
     a.hist_pool0 = (uint32_t*)hist_pool0.data_ptr<int>();
     a.hist_pool1 = (uint32_t*)hist_pool1.data_ptr<int>();
-    a.pool_count = pool_count.data_ptr<int>();
     a.pool_cap = (int)pool_cap;
     a.hist_save_min = HIST_SAVE_MIN;
     a.small = (WorkItem*)small_q.data_ptr();
-    a.small_count = small_count.data_ptr<int>();
     a.small_cap = (int)(S + 2);
     a.mid = (WorkItem*)mid_q.data_ptr();
-    a.mid_count = mid_count.data_ptr<int>();
     a.mid_cap = (int)mid_cap;
+    int* st = state.data_ptr<int>();
+    a.pool_count = st + 1;   // kernel indexes [wp * 4]
 
     int GRID = 4096;
     if (const char* e = getenv("FLAKE16_FIT_GRID")) GRID = atoi(e);
@@ -161,21 +162,16 @@ std::vector<at::Tensor> forest_fit(
     while (!done) {
         for (int c = 0; c < CHUNK; ++c, ++lev) {
             int nx = cur ^ 1;
-            CHECK_HIP(hipMemsetAsync(counts.data_ptr<int>() + nx, 0, 4,
-                                     stream));
-            // reset the write-parity histogram-pool counter for this level
-            CHECK_HIP(hipMemsetAsync(
-                pool_count.data_ptr<int>() + ((lev + 1) & 1), 0, 4, stream));
+            // one 16 B memset clears next-level work/pool/small/mid counts
+            CHECK_HIP(hipMemsetAsync(st + nx * 4, 0, 16, stream));
             a.sidx_cur = (cur == 0 ? sidx_a : sidx_b).data_ptr<int>();
             a.sidx_nxt = (cur == 0 ? sidx_b : sidx_a).data_ptr<int>();
             a.cur = (const WorkItem*)(cur == 0 ? work_a : work_b).data_ptr();
             a.nxt = (WorkItem*)(cur == 0 ? work_b : work_a).data_ptr();
-            a.cur_count = counts.data_ptr<int>() + cur;
-            a.nxt_count = counts.data_ptr<int>() + nx;
-            CHECK_HIP(hipMemsetAsync(small_count.data_ptr<int>(), 0, 4,
-                                     stream));
-            CHECK_HIP(hipMemsetAsync(mid_count.data_ptr<int>(), 0, 4,
-                                     stream));
+            a.cur_count = st + cur * 4;
+            a.nxt_count = st + nx * 4;
+            a.small_count = st + nx * 4 + 2;
+            a.mid_count = st + nx * 4 + 3;
             if (splitter_random) {
                 et_split_kernel<<<GRID, HBLK, 0, stream>>>(a);
             } else if (!has_wide && max_features < F &&
@@ -192,7 +188,7 @@ std::vector<at::Tensor> forest_fit(
             small_subtree_kernel<<<2048, HBLK, 0, stream>>>(
                 a, a.sidx_nxt);
             CHECK_HIP(hipMemcpyAsync(pinned_p + (c % PINSZ),
-                                     counts.data_ptr<int>() + nx, 4,
+                                     st + nx * 4, 4,
                                      hipMemcpyDeviceToHost, stream));
             cur = nx;
         }
