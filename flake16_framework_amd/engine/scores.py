@@ -13,7 +13,8 @@ Execution model (MI355X-first, nothing like the reference's process pool):
   cells device-resident — fold-batched balancing, binning, forest fit,
   prediction and confusion — and the per-cell result blobs are combined with
   one RCCL all-reduce at the end (parallel/comm.py).  The 'ref' backend runs
-  the identical algorithm in numpy on CPU.
+  the identical algorithm in numpy on CPU, optionally over a process pool
+  (the reference's execution model).
 """
 
 import pickle
@@ -103,11 +104,14 @@ def evaluate_cell_ref(config_keys, cell_idx, tests=None, tests_file=None,
 
 
 def run_scores(tests_file=None, tests=None, backend="auto", cells=None,
-               progress=None, seed=GLOBAL_SEED, on_result=None):
+               progress=None, seed=GLOBAL_SEED, on_result=None,
+               processes=1):
     """Evaluate `cells` (an iterable of (cell_idx, config_keys); default all
     216) and return {config_keys: [t_train, t_test, scores, scores_total]}.
     on_result(config_keys, value): optional per-cell callback (used by the
-    checkpoint writer).
+    checkpoint writer).  processes > 1 runs the 'ref' backend cells in a
+    multiprocessing pool (the reference's execution model,
+    experiment.py:496-498); the hip backend uses streams instead.
     """
     from ..utils.trace import trace_span
 
@@ -125,6 +129,22 @@ def run_scores(tests_file=None, tests=None, backend="auto", cells=None,
 
     out = {}
     t_start = time.time()
+
+    if processes > 1:
+        from multiprocessing import Pool
+        args = [(keys, ci, tests, tests_file, seed)
+                for ci, keys in all_cells]
+        with Pool(processes=processes) as pool:
+            for n_done, (keys, value) in enumerate(
+                    pool.imap_unordered(_eval_cell_ref_task, args)):
+                out[keys] = value
+                if on_result:
+                    on_result(keys, value)
+                if progress:
+                    progress(n_done + 1, len(all_cells),
+                             time.time() - t_start, ", ".join(keys))
+        return out
+
     for n_done, (cell_idx, config_keys) in enumerate(all_cells):
         with trace_span("cell", cell=cell_idx, backend="ref"):
             out[config_keys] = evaluate_cell_ref(
@@ -136,6 +156,12 @@ def run_scores(tests_file=None, tests=None, backend="auto", cells=None,
             progress(n_done + 1, len(all_cells), time.time() - t_start,
                      ", ".join(config_keys))
     return out
+
+
+def _eval_cell_ref_task(args):
+    keys, cell_idx, tests, tests_file, seed = args
+    return keys, evaluate_cell_ref(keys, cell_idx, tests=tests,
+                                   tests_file=tests_file, seed=seed)
 
 
 def _run_scores_hip(all_cells, tests, tests_file, seed, progress,
@@ -254,10 +280,13 @@ def write_scores(tests_file=None, scores_file=SCORES_FILE, backend="auto",
         print(f"[rank {rank}] {done_n}/{total} {name} "
               f"({elapsed:.0f}s elapsed, eta {eta:.0f}s)", flush=True)
 
+    import os as _os
+    n_proc = int(_os.environ.get("FLAKE16_REF_PROCS",
+                                 str(_os.cpu_count() or 1)))
     with trace_span("scores_sweep", rank_cells=len(my_cells)):
         result = run_scores(tests_file=tests_file, backend=backend,
                             cells=my_cells, progress=progress,
-                            on_result=on_result)
+                            on_result=on_result, processes=n_proc)
     result.update(done)
     result = comm.gather_scores(result)
 
