@@ -1,15 +1,23 @@
 // Batched histogram-forest construction for MI355X (gfx950).
 //
 // One forest_fit call builds ALL trees of one grid cell (10 folds x
-// n_estimators jobs) device-resident: a level-synchronous work queue of
-// (job, node) items, one 256-thread workgroup per active node, per-node
-// LDS histograms (16 features x 256 bins x {total, class1} = 32 KiB),
-// split selection in fp64 (file compiled -ffp-contract=off so scores are
-// bit-identical to the numpy reference in models/forest_ref.py), and a
-// stable in-kernel partition into the next level's sample-index buffer.
-//
-// All randomness is Philox keyed on (tag, node sample-range, draw), so the
-// nondeterministic node-allocation order (atomics) cannot perturb trees.
+// n_estimators jobs) device-resident.  Work is tiered by node size:
+//   > 2048 samples   level-synchronous work queue, one 256-thread
+//                    workgroup per node, packed 16-bit LDS histograms
+//                    (hist_split_kernel; a WIDE unpacked instantiation
+//                    covers nodes >= 2^16) with parent-minus-smaller-child
+//                    histogram-subtraction pools; Extra-Trees jobs use the
+//                    histogram-free two-pass et_split_kernel instead
+//   65..2048         mid_subtree_kernel: the block stages the node's code
+//                    rows into LDS once and finishes the whole subtree
+//                    internally (LDS histograms + LDS index partitions)
+//   <= 64            small_subtree_kernel: one 64-lane wave finishes the
+//                    subtree from register-resident samples with
+//                    ballot/popcount counting
+// Split selection is fp64 (file compiled -ffp-contract=off) and all
+// randomness is Philox keyed on (tag, node sample-range, draw), so trees
+// are bit-identical to the numpy reference in models/forest_ref.py no
+// matter how the device schedules the work.
 //
 // Reference semantics being implemented: sklearn 1.0.2 defaults for
 // DecisionTree/RandomForest/ExtraTrees (see models/forest_ref.py docstring;
