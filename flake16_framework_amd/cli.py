@@ -54,6 +54,7 @@ def main(argv=None):
     elif command == "scores":
         backend = _pop_flag(args, "--backend", "auto")
         checkpoint = _pop_flag(args, "--checkpoint", None)
+        n_cells = _pop_flag(args, "--cells", None)
         trace = _pop_flag(args, "--trace", None)
         if trace:
             from .utils.trace import set_trace_file
@@ -61,7 +62,8 @@ def main(argv=None):
         from .engine.scores import write_scores
         from .parallel import comm
         comm.init_from_env()
-        write_scores(backend=backend, checkpoint=checkpoint)
+        write_scores(backend=backend, checkpoint=checkpoint,
+                     n_cells=int(n_cells) if n_cells else None)
     elif command == "shap":
         backend = _pop_flag(args, "--backend", "auto")
         from .engine.shap_stage import write_shap

@@ -245,7 +245,7 @@ def _load_checkpoint(path):
 
 
 def write_scores(tests_file=None, scores_file=SCORES_FILE, backend="auto",
-                 checkpoint=None):
+                 checkpoint=None, n_cells=None):
     """Full 216-cell sweep (sharded across ranks if distributed is
     initialized) -> scores.pkl on rank 0.
 
@@ -258,7 +258,7 @@ def write_scores(tests_file=None, scores_file=SCORES_FILE, backend="auto",
     from ..utils.trace import trace_span
 
     rank, world = comm.rank_world()
-    my_cells = comm.shard_cells(world, rank)
+    my_cells = comm.shard_cells(world, rank, n_cells=n_cells)
 
     done = {}
     on_result = None

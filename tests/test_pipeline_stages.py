@@ -333,3 +333,41 @@ class TestChurnCollector:
         churn = collect_churn(str(tmp_path), None)
         assert churn["a.py"][2] >= 2   # line 2 touched by both commits
         assert churn["a.py"][1] >= 1
+
+
+def _pool_task(x):
+    import time
+    time.sleep(0.01)
+    return f"done: {x}", x * 2
+
+
+class TestManagePool:
+    def test_progress_and_results(self, capsys):
+        from multiprocessing import Pool
+        from flake16_framework_amd.orchestrate.runner import manage_pool
+        with Pool(2) as pool:
+            results = sorted(manage_pool(pool, _pool_task, [1, 2, 3]))
+        assert results == [2, 4, 6]
+        out = capsys.readouterr().out
+        assert "done:" in out and "3/0" in out
+
+
+class TestScoresCli:
+    def test_scores_cli_ref_backend_with_cells(self, tmp_path):
+        import pickle
+        env = dict(os.environ, PYTHONPATH=REPO, FLAKE16_REF_PROCS="2")
+        run = lambda *args: subprocess.run(
+            [sys.executable, os.path.join(REPO, "experiment.py"), *args],
+            cwd=str(tmp_path), env=env, capture_output=True, text=True,
+            timeout=600)
+        r = run("synthetic", "--n-tests", "400", "--seed", "2")
+        assert r.returncode == 0, r.stderr
+        r = run("scores", "--backend", "ref", "--cells", "3",
+                "--checkpoint", "ck")
+        assert r.returncode == 0, r.stderr
+        with open(tmp_path / "scores.pkl", "rb") as fd:
+            scores = pickle.load(fd)
+        assert len(scores) == 3
+        for keys, val in scores.items():
+            assert len(keys) == 5 and len(val) == 4
+        assert (tmp_path / "ck.rank0").exists()
