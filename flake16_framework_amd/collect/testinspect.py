@@ -35,6 +35,8 @@ import subprocess
 import sys
 import time
 
+from pytest import hookimpl as _pytest_hookimpl
+
 
 def pytest_addoption(parser):
     group = parser.getgroup("testinspect")
@@ -204,6 +206,10 @@ class TestInspectPlugin:
             name = item.name.split("[")[0]
             self.test_items[item.nodeid] = (path, name)
 
+    # hookwrapper: the default pytest_runtest_call runs the test inside
+    # our tracing (a plain hookimpl that called item.runtest() itself
+    # would EXECUTE THE TEST TWICE — the default impl still runs).
+    @_pytest_hookimpl(hookwrapper=True)
     def pytest_runtest_call(self, item):
         proc = None
         try:
@@ -219,7 +225,7 @@ class TestInspectPlugin:
         old = sys.gettrace()
         sys.settrace(tracer)
         try:
-            item.runtest()
+            yield
         finally:
             sys.settrace(old)
             t_exec = time.time() - t0

@@ -21,8 +21,6 @@ SUITE = '''
 import os
 import sys
 
-MARKER = os.path.join(os.path.dirname(__file__), "breaker_ran")
-
 
 def _run_info():
     # container name encodes (mode, run number): proj_<mode>_<runN>
@@ -35,6 +33,12 @@ def _run_info():
 
 
 MODE, RUN_N = _run_info()
+
+# per-run marker: local-mode runs share the checkout (real study runs are
+# container-isolated), so order-dependence state must not leak across the
+# concurrent pool runs
+MARKER = os.path.join(os.path.dirname(__file__),
+                      "breaker_ran_%s_%d" % (MODE, RUN_N))
 
 if os.path.exists(MARKER):
     os.remove(MARKER)

@@ -371,3 +371,28 @@ class TestScoresCli:
         for keys, val in scores.items():
             assert len(keys) == 5 and len(val) == 4
         assert (tmp_path / "ck.rank0").exists()
+
+
+class TestSingleExecution:
+    def test_testinspect_runs_each_test_once(self, tmp_path):
+        """Regression: a plain pytest_runtest_call hookimpl that invokes
+        item.runtest() itself runs every test twice (the default impl
+        still fires) — the collector must be a hookwrapper."""
+        suite = tmp_path / "test_count.py"
+        counter = tmp_path / "count.txt"
+        suite.write_text(f'''
+def test_once():
+    p = {str(counter)!r}
+    n = int(open(p).read()) if __import__("os").path.exists(p) else 0
+    with open(p, "w") as fd:
+        fd.write(str(n + 1))
+''')
+        env = dict(os.environ, PYTHONPATH=REPO)
+        proc = subprocess.run(
+            [sys.executable, "-m", "pytest", "-q", str(suite),
+             "-p", "flake16_framework_amd.collect.testinspect",
+             f"--testinspect={tmp_path}/ti", "--rootdir", str(tmp_path)],
+            cwd=str(tmp_path), env=env, capture_output=True, text=True,
+            timeout=120)
+        assert proc.returncode == 0, proc.stdout + proc.stderr
+        assert counter.read_text() == "1"
