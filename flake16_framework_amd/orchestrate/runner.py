@@ -164,9 +164,12 @@ def iter_containers(run_modes, subjects_file=SUBJECTS_FILE, n_runs=None):
                 yield f"{proj}_{mode}_{run_n}", commands
 
 
-def manage_pool(pool, fn, args, out=sys.stdout):
+def manage_pool(pool, fn, args, out=None):
     """Shuffle, imap_unordered, and print per-task progress lines with
-    elapsed/ETA minutes — the reference's pool meter (experiment.py:191)."""
+    elapsed/ETA minutes — the reference's pool meter (experiment.py:191).
+    `out` resolves to the CURRENT sys.stdout at call time (an import-time
+    default would capture a stale stream under redirection)."""
+    out = out or sys.stdout
     n_finish = 0
     t_start = time.time()
 
