@@ -100,8 +100,10 @@ std::vector<at::Tensor> forest_fit(
     int HIST_SAVE_MIN = 2048;
     if (const char* e = getenv("FLAKE16_HIST_SAVE_MIN"))
         HIST_SAVE_MIN = atoi(e);
+    // cap bounds pool memory at ~2 GB per parity (16 KiB per slot);
+    // exhaustion is correct but silently degrades to full accumulation
     const long pool_cap =
-        std::min<long>(2 * (S / HIST_SAVE_MIN) + 8, 32768);
+        std::min<long>(2 * (S / HIST_SAVE_MIN) + 8, 131072);
     auto hist_pool0 = at::empty({pool_cap * FPAD * 256},
                                 codes.options().dtype(at::kInt));
     auto hist_pool1 = at::empty({pool_cap * FPAD * 256},
