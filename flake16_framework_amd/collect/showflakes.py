@@ -36,11 +36,17 @@ def pytest_collection_modifyitems(config, items):
 
 
 class _Recorder:
+    """Registered as a per-session pytest plugin only when --record-file is
+    active, so runs without the option pay nothing and two sessions in one
+    process cannot leak reports into each other."""
+
     def __init__(self, path):
         self.path = path
         self.outcomes = {}
 
-    def record(self, report):
+    def pytest_runtest_logreport(self, report):
+        if report.when not in ("setup", "call", "teardown"):
+            return
         nid = report.nodeid
         prev = self.outcomes.get(nid)
         if report.outcome == "failed" or prev == "failed":
@@ -51,7 +57,7 @@ class _Recorder:
             outcome = "passed"
         self.outcomes[nid] = outcome
 
-    def flush(self):
+    def pytest_sessionfinish(self, session, exitstatus):
         with open(self.path, "w") as fd:
             for nid, outcome in self.outcomes.items():
                 fd.write(f"{outcome}\t{nid}\n")
@@ -60,30 +66,20 @@ class _Recorder:
 def pytest_configure(config):
     path = config.getoption("--record-file")
     if path:
-        config._showflakes_recorder = _Recorder(path)
+        rec = _Recorder(path)
+        config._showflakes_recorder = rec
+        config.pluginmanager.register(rec)
 
 
-def pytest_runtest_logreport(report):
-    import pytest  # noqa: F401
-    # recorder reached via the config on the session; stored at configure
-    # time — pytest passes report without config, so use the plugin trick:
-    _report_sink.append(report)
-
-
-_report_sink = []
+def pytest_unconfigure(config):
+    rec = getattr(config, "_showflakes_recorder", None)
+    if rec is not None:
+        config.pluginmanager.unregister(rec)
+        del config._showflakes_recorder
 
 
 def pytest_sessionfinish(session, exitstatus):
-    config = session.config
-    rec = getattr(config, "_showflakes_recorder", None)
-    if rec is not None:
-        for report in _report_sink:
-            if report.when in ("setup", "call", "teardown"):
-                rec.record(report)
-        rec.flush()
-    _report_sink.clear()
-
-    if config.getoption("--set-exitstatus"):
+    if session.config.getoption("--set-exitstatus"):
         # ordinary test failures are data, not an orchestration error
         if exitstatus == 1:  # pytest.ExitCode.TESTS_FAILED
             session.exitstatus = 0

@@ -266,6 +266,12 @@ def write_scores(tests_file=None, scores_file=SCORES_FILE, backend="auto",
         ckpt_path = f"{checkpoint}.rank{rank}"
         done = _load_checkpoint(ckpt_path)
         all_keys = list(iter_config_keys())
+        # Keep only entries this rank OWNS under the current shard: a resume
+        # with a different world size would otherwise merge a cell from the
+        # old shard AND recompute it on its new owner, and the all-reduce SUM
+        # in gather_scores would double its counts.
+        my_keys = {all_keys[c] for c in my_cells}
+        done = {k: v for k, v in done.items() if k in my_keys}
         my_cells = [c for c in my_cells if all_keys[c] not in done]
         ckpt_fd = open(ckpt_path, "ab")
         ckpt_lock = __import__("threading").Lock()

@@ -34,6 +34,12 @@ PIP_INSTALL = ["pip", "install", "-I", "--no-deps"]
 COLLECT_PLUGINS = ("flake16_framework_amd.collect.showflakes",
                    "flake16_framework_amd.collect.testinspect")
 
+# Root of the framework checkout (the directory holding the
+# flake16_framework_amd package) — exposed to subject venvs by
+# install_plugins so `pytest -p flake16_framework_amd.collect.*` resolves.
+FRAMEWORK_ROOT = os.path.dirname(os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
+
 
 def iter_subjects(subjects_file=SUBJECTS_FILE):
     """subjects.txt rows: owner/repo,sha,package_dir[,pre-command...]."""
@@ -64,6 +70,25 @@ def setup_project(proj, url, sha, package_dir, subjects_dir=SUBJECTS_DIR):
     sp.run([*PIP_INSTALL, PIP_VERSION], env=env, check=True)
     sp.run([*PIP_INSTALL, "-r", requirements_file], env=env, check=True)
     sp.run([*PIP_INSTALL, "-e", package_dir], env=env, check=True)
+    install_plugins(venv_dir)
+
+
+def install_plugins(venv_dir, framework_root=FRAMEWORK_ROOT):
+    """Make the collection plugins importable from the subject venv (the
+    reference pip-installs its showflakes/testinspect checkouts into each
+    venv, experiment.py:125).  The framework's own setup.py builds the HIP
+    extension and needs torch, which subject venvs don't ship, so the
+    package is exposed with a .pth in the venv's site-packages instead of
+    a pip install — equivalent for `pytest -p flake16_framework_amd...`."""
+    import glob
+    site_dirs = glob.glob(os.path.join(venv_dir, "lib", "python*",
+                                       "site-packages"))
+    if not site_dirs:
+        raise RuntimeError(f"no site-packages under {venv_dir}")
+    for site_dir in site_dirs:
+        pth = os.path.join(site_dir, "flake16_framework_amd.pth")
+        with open(pth, "w") as fd:
+            fd.write(framework_root + "\n")
 
 
 def setup_image():
