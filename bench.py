@@ -105,7 +105,7 @@ def main():
     if world > 1:
         import torch.distributed as dist
         t = torch.tensor([elapsed], dtype=torch.float64,
-                         device="cuda" if use_cuda else "cpu")
+                         device=comm.collective_device())
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.cpu().item())
 

@@ -88,6 +88,10 @@ struct ForestDev {
     WorkItem* __restrict__ mid;
     int* __restrict__ mid_count;
     int mid_cap;
+    // mid_subtree_kernel mode: wave-parallel node processing (4 nodes in
+    // flight per block) when max_features <= WAVE_CANDS; 0 = block-serial
+    // DFS (DT, or FLAKE16_NO_WAVE_MID=1 for A/B runs).
+    int wave_mid;
 };
 
 #define SMALL_N 64
@@ -1106,7 +1110,303 @@ struct MidFrame {
     int node;
 };
 
-#define MID_STACK 16   // smaller-child-first: <= log2(MID_N)+2
+// Block-serial DFS (DT / fallback) needs <= log2(MID_N)+2 entries; the
+// wave-parallel mode (4 nodes in flight, breadth-ish order) can hold the
+// whole >SMALL_N frontier of a MID_N subtree (~ MID_N/SMALL_N * 2, plus
+// slack); overflow falls back to the next-level global work queue.
+#define MID_STACK 96
+#define WAVE_CANDS 4   // wave-parallel path: max_features <= 4 (RF / ET)
+
+// One 64-lane wave splits one node [ls, le) of the staged window and
+// routes its children (shared LDS stack for > SMALL_N, global small queue
+// otherwise).  Control flow is wave-uniform throughout; split semantics,
+// Philox counters (global job-relative ranges) and fp64 scores are
+// identical to the block-wide path — trees are bit-identical.  Requires
+// max_features <= WAVE_CANDS (RF / ET; DT keeps the block path).
+// whist: this wave's WAVE_CANDS*256-word LDS histogram region (RF only).
+__device__ __forceinline__ void mid_wave_node(
+    const ForestDev& a, const WorkItem& it, int ls, int le, int depth,
+    int node, const uint4* m_codes, const uint8_t* m_lab, uint16_t* m_idx,
+    uint16_t* m_idx2, uint32_t* whist, int* wperm, uint32_t* wthr,
+    int* wcand, int* wcbin, MidFrame* stack, int* sh_count) {
+    const int lane = threadIdx.x & 63;
+    const int n = le - ls;
+    const long nbase = a.j_node_off[it.job];
+    const uint32_t key = (uint32_t)a.j_key[it.job];
+    const int F = a.F;
+    const int gs = it.start + ls;   // global job-relative range (RNG id)
+    const int ge = it.start + le;
+
+    // Pass 1: per-feature occupied min/max + class-1 count.
+    int lmin[FPAD], lmax[FPAD];
+    #pragma unroll
+    for (int f = 0; f < FPAD; ++f) { lmin[f] = 256; lmax[f] = -1; }
+    int lc1 = 0;
+    for (int i = ls + lane; i < le; i += 64) {
+        const int o = m_idx[i];
+        const uint4 cw = m_codes[o];
+        const uint32_t w[4] = {cw.x, cw.y, cw.z, cw.w};
+        lc1 += m_lab[o];
+        #pragma unroll
+        for (int f = 0; f < FPAD; ++f) {
+            const int b = (int)((w[f >> 2] >> ((f & 3) * 8)) & 0xFFu);
+            lmin[f] = min(lmin[f], b);
+            lmax[f] = max(lmax[f], b);
+        }
+    }
+    for (int d = 32; d > 0; d >>= 1) lc1 += __shfl_down(lc1, d);
+    const int c1 = __shfl(lc1, 0);
+    const int c0 = n - c1;
+    #pragma unroll
+    for (int f = 0; f < FPAD; ++f) {
+        int mn = lmin[f], mx = lmax[f];
+        for (int d = 32; d > 0; d >>= 1) {
+            mn = min(mn, __shfl_xor(mn, d));
+            mx = max(mx, __shfl_xor(mx, d));
+        }
+        lmin[f] = mn;   // wave-uniform from here on
+        lmax[f] = mx;
+    }
+
+    if (lane == 0) {
+        a.ncnt0[nbase + node] = (float)c0;
+        a.ncnt1[nbase + node] = (float)c1;
+    }
+    if (n < 2 || c0 == 0 || c1 == 0) return;   // leaf
+
+    // Feature permutation + candidate walk (draws parallel across lanes,
+    // walk on lane 0 through this wave's LDS scratch).
+    {
+        const uint32_t uf = philox_draw(
+            TAG_FEATSEL | ((uint32_t)(depth & 0xFF) << 8),
+            (uint32_t)gs, (uint32_t)ge,
+            (uint32_t)(lane < FPAD ? lane : 0), a.seed, key);
+        if (a.splitter_random && lane < FPAD)
+            wthr[lane] = philox_draw(
+                TAG_THRESH | ((uint32_t)(depth & 0xFF) << 8),
+                (uint32_t)gs, (uint32_t)ge, (uint32_t)lane, a.seed, key);
+        if (lane == 0)
+            for (int f = 0; f < F; ++f) wperm[f] = f;
+        for (int i = 0; i < F - 1; ++i) {
+            const uint32_t u = __shfl(uf, i);
+            if (lane == 0) {
+                int j = i + (int)philox_bounded(u, (uint32_t)(F - i));
+                int t = wperm[i]; wperm[i] = wperm[j]; wperm[j] = t;
+            }
+        }
+    }
+    int ncand = 0;
+    if (lane == 0) {
+        for (int i = 0; i < F && ncand < a.max_features; ++i) {
+            const int f = wperm[i];
+            if (lmin[f] == lmax[f]) continue;
+            wcand[ncand] = f;
+            if (a.splitter_random)
+                wcbin[ncand] = lmin[f] + (int)philox_bounded(
+                    wthr[f], (uint32_t)(lmax[f] - lmin[f]));
+            ++ncand;
+        }
+    }
+    ncand = __shfl(ncand, 0);
+    if (ncand == 0) return;   // all candidates constant: leaf
+
+    double best_s = -1.0e300;
+    int bf = -1, bb = -1, bnl = 0;
+
+    if (a.splitter_random) {
+        // ET: counts at each candidate's drawn bin — histogram-free.
+        int cnt[WAVE_CANDS], cnt1[WAVE_CANDS];
+        #pragma unroll
+        for (int ci = 0; ci < WAVE_CANDS; ++ci) { cnt[ci] = 0; cnt1[ci] = 0; }
+        for (int i = ls + lane; i < le; i += 64) {
+            const int o = m_idx[i];
+            const uint4 cw = m_codes[o];
+            const uint32_t w[4] = {cw.x, cw.y, cw.z, cw.w};
+            const int lab = m_lab[o];
+            #pragma unroll
+            for (int ci = 0; ci < WAVE_CANDS; ++ci) {
+                if (ci >= ncand) break;
+                const int f = wcand[ci];
+                const int b = (int)((w[f >> 2] >> ((f & 3) * 8)) & 0xFFu);
+                const int le_ = b <= wcbin[ci];
+                cnt[ci] += le_;
+                cnt1[ci] += le_ & lab;
+            }
+        }
+        #pragma unroll
+        for (int ci = 0; ci < WAVE_CANDS; ++ci) {
+            if (ci >= ncand) break;
+            int c = cnt[ci], cl = cnt1[ci];
+            for (int d = 32; d > 0; d >>= 1) {
+                c += __shfl_down(c, d);
+                cl += __shfl_down(cl, d);
+            }
+            c = __shfl(c, 0);
+            cl = __shfl(cl, 0);
+            const long nL = c, n1L = cl;
+            const long n0L = nL - n1L, nR = n - nL;
+            const long n1R = c1 - n1L, n0R = c0 - n0L;
+            if (nL == 0 || nR == 0) continue;
+            const double sc = (double)(n0L * n0L + n1L * n1L) / (double)nL
+                            + (double)(n0R * n0R + n1R * n1R) / (double)nR;
+            if (sc > best_s) {
+                best_s = sc;
+                bf = wcand[ci];
+                bb = wcbin[ci];
+                bnl = (int)nL;
+            }
+        }
+    } else {
+        // RF: candidate-only packed LDS histograms.
+        for (int i = lane; i < ncand * 256; i += 64) whist[i] = 0;
+        for (int i = ls + lane; i < le; i += 64) {
+            const int o = m_idx[i];
+            const uint4 cw = m_codes[o];
+            const uint32_t w[4] = {cw.x, cw.y, cw.z, cw.w};
+            const uint32_t inc = 1u | ((uint32_t)m_lab[o] << 16);
+            #pragma unroll
+            for (int ci = 0; ci < WAVE_CANDS; ++ci) {
+                if (ci >= ncand) break;
+                const int f = wcand[ci];
+                const uint32_t b = (w[f >> 2] >> ((f & 3) * 8)) & 0xFFu;
+                atomicAdd(&whist[ci * 256 + b], inc);
+            }
+        }
+        for (int ci = 0; ci < ncand; ++ci) {
+            const int f = wcand[ci];
+            const int bmin = lmin[f], bmax = lmax[f];
+            int ln[4], l1[4];
+            int tn = 0, t1 = 0;
+            #pragma unroll
+            for (int k = 0; k < 4; ++k) {
+                const uint32_t v = whist[ci * 256 + lane * 4 + k];
+                ln[k] = (int)(v & 0xFFFFu);
+                l1[k] = (int)(v >> 16);
+                tn += ln[k];
+                t1 += l1[k];
+            }
+            int scn = tn, sc1 = t1;
+            for (int d = 1; d < 64; d <<= 1) {
+                const int un = __shfl_up(scn, d);
+                const int u1 = __shfl_up(sc1, d);
+                if (lane >= d) { scn += un; sc1 += u1; }
+            }
+            int cn = scn - tn, c1f = sc1 - t1;
+            double cand_s = -1.0;
+            int cand_b = -1, cand_nl = 0;
+            #pragma unroll
+            for (int k = 0; k < 4; ++k) {
+                const int b = lane * 4 + k;
+                cn += ln[k];
+                c1f += l1[k];
+                if (ln[k] == 0 || b < bmin || b >= bmax) continue;
+                const long nL = cn, n1L = c1f;
+                const long n0L = nL - n1L, nR = n - nL;
+                const long n1R = c1 - n1L, n0R = c0 - n0L;
+                const double s =
+                    (double)(n0L * n0L + n1L * n1L) / (double)nL
+                    + (double)(n0R * n0R + n1R * n1R) / (double)nR;
+                if (s > cand_s) { cand_s = s; cand_b = b; cand_nl = (int)nL; }
+            }
+            for (int d = 32; d > 0; d >>= 1) {
+                const double os = __shfl_down(cand_s, d);
+                const int ob = __shfl_down(cand_b, d);
+                const int onl = __shfl_down(cand_nl, d);
+                if (os > cand_s || (os == cand_s && ob != -1 &&
+                                    (cand_b == -1 || ob < cand_b))) {
+                    cand_s = os; cand_b = ob; cand_nl = onl;
+                }
+            }
+            cand_s = __shfl(cand_s, 0);
+            cand_b = __shfl(cand_b, 0);
+            cand_nl = __shfl(cand_nl, 0);
+            // sequential select in perm order, strict > (phase-6 contract)
+            if (cand_b >= 0 && cand_s > best_s) {
+                best_s = cand_s;
+                bf = wcand[ci];
+                bb = cand_b;
+                bnl = cand_nl;
+            }
+        }
+    }
+
+    if (bf < 0) return;   // no valid split: leaf
+    const int nL = bnl;
+
+    int lid = 0;
+    if (lane == 0) {
+        lid = atomicAdd(&a.node_alloc[it.job], 2);
+        a.nfeat[nbase + node] = bf;
+        a.nsplit[nbase + node] = bb;
+        a.nleft[nbase + node] = lid;
+    }
+    lid = __shfl(lid, 0);
+
+    // Stable partition of m_idx[ls, le) via this node's private m_idx2
+    // range (sibling ranges are disjoint — no cross-wave aliasing).
+    int lo = 0, ro = 0;
+    for (int base = ls; base < le; base += 64) {
+        const int i = base + lane;
+        const bool valid = i < le;
+        int o = 0, flag = 0;
+        if (valid) {
+            o = m_idx[i];
+            const uint4 cw = m_codes[o];
+            const uint32_t wsel = (&cw.x)[bf >> 2];
+            flag = (int)(((wsel >> ((bf & 3) * 8)) & 0xFFu)
+                         <= (uint32_t)bb);
+        }
+        const unsigned long long lm = __ballot(valid && flag);
+        const unsigned long long below = (1ULL << lane) - 1ULL;
+        const int rank = __popcll(lm & below);
+        const int chunk_left = __popcll(lm);
+        const int chunk_n = min(64, le - base);
+        if (valid) {
+            const int dst = flag
+                ? ls + lo + rank
+                : ls + nL + ro + (i - base) - rank;
+            m_idx2[dst] = (uint16_t)o;
+        }
+        lo += chunk_left;
+        ro += chunk_n - chunk_left;
+    }
+    for (int i = ls + lane; i < le; i += 64) m_idx[i] = m_idx2[i];
+
+    // Route children: > SMALL_N to the shared stack (overflow: next-level
+    // global work queue — correct, just slower); <= SMALL_N to the global
+    // small queue with final global ranges.
+    if (lane == 0) {
+        const MidFrame kids[2] = {
+            {(short)ls, (short)(ls + nL), depth + 1, lid},
+            {(short)(ls + nL), (short)le, depth + 1, lid + 1},
+        };
+        for (int x = 0; x < 2; ++x) {
+            const MidFrame& fr = kids[x];
+            const int fn = fr.e - fr.s;
+            if (fn > SMALL_N) {
+                const int idx = atomicAdd(sh_count, 1);
+                if (idx < MID_STACK) {
+                    stack[idx] = fr;
+                } else {
+                    atomicSub(sh_count, 1);
+                    const int i2 = atomicAdd(a.nxt_count, 1);
+                    if (i2 < a.work_cap)
+                        a.nxt[i2] = {it.job, fr.node, it.start + fr.s,
+                                     it.start + fr.e, fr.depth, -1};
+                    else
+                        atomicExch(a.err_flag, 1);
+                }
+            } else {
+                const int i2 = atomicAdd(a.small_count, 1);
+                if (i2 < a.small_cap)
+                    a.small[i2] = {it.job, fr.node, it.start + fr.s,
+                                   it.start + fr.e, fr.depth, -1};
+                else
+                    atomicExch(a.err_flag, 1);
+            }
+        }
+    }
+}
 
 __launch_bounds__(HBLK)
 __global__ void mid_subtree_kernel(ForestDev a,
@@ -1127,6 +1427,13 @@ __global__ void mid_subtree_kernel(ForestDev a,
     __shared__ MidFrame stack[MID_STACK];
     __shared__ int sh_sp;
     __shared__ int sh_lo, sh_ro;
+    // wave-parallel mode scratch (one slice per wave)
+    __shared__ MidFrame wframes[HBLK / 64];
+    __shared__ int wperm_ws[HBLK / 64][FPAD];
+    __shared__ uint32_t wthr_ws[HBLK / 64][FPAD];
+    __shared__ int wcand_ws[HBLK / 64][WAVE_CANDS];
+    __shared__ int wcbin_ws[HBLK / 64][WAVE_CANDS];
+    __shared__ int sh_count, sh_take;
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -1151,6 +1458,44 @@ __global__ void mid_subtree_kernel(ForestDev a,
             m_lab[i] = a.labels[row];
             m_idx[i] = (uint16_t)i;
         }
+
+        if (a.wave_mid) {
+            // Wave-parallel rounds: up to HBLK/64 nodes in flight, each
+            // built by one wave (no block barriers inside a node).  The
+            // block-wide 16 KiB `hist` array is re-sliced as one
+            // WAVE_CANDS*256-word histogram region per wave.
+            if (tid == 0) {
+                sh_count = 1;
+                stack[0] = {0, (short)n0, it.depth, it.node};
+            }
+            while (true) {
+                __syncthreads();
+                if (tid == 0) {
+                    const int take = min(HBLK / 64, sh_count);
+                    for (int w = 0; w < take; ++w)
+                        wframes[w] = stack[sh_count - 1 - w];
+                    sh_count -= take;
+                    sh_take = take;
+                }
+                __syncthreads();
+                if (sh_take == 0) break;
+                if (wave < sh_take) {
+                    const MidFrame fr = wframes[wave];
+                    mid_wave_node(a, it, fr.s, fr.e, fr.depth, fr.node,
+                                  m_codes, m_lab, m_idx, m_idx2,
+                                  hist + wave * (WAVE_CANDS * 256),
+                                  wperm_ws[wave], wthr_ws[wave],
+                                  wcand_ws[wave], wcbin_ws[wave],
+                                  stack, &sh_count);
+                }
+            }
+            __syncthreads();
+            for (int i = tid; i < n0; i += HBLK)
+                a.sidx_nxt[sbase + it.start + i] = m_rows[m_idx[i]];
+            __syncthreads();
+            continue;
+        }
+
         if (tid == 0) {
             sh_sp = 0;
             stack[0] = {0, (short)n0, it.depth, it.node};
@@ -1614,15 +1959,25 @@ __global__ void small_subtree_kernel(ForestDev a,
                     // code as the threshold candidate (duplicates give
                     // identical (score, bin) pairs; the argmax tie-break
                     // to the lowest bin matches the ascending sequential
-                    // scan exactly)
-                    int cnt = 0, cnt1 = 0;
-                    for (int d = 0; d < 64; ++d) {
-                        const int cd = __shfl(my_code, d);
-                        const bool le = ((mask >> d) & 1ULL) &&
-                                        cd <= my_code;
-                        cnt += le;
-                        cnt1 += le && ((lab_mask >> d) & 1ULL);
+                    // scan exactly).  The rank (lanes with code <= mine)
+                    // comes from 8 bit-plane ballots instead of a 64-step
+                    // shuffle scan: after the loop E = lanes whose code
+                    // equals mine, L = lanes whose code is strictly less.
+                    unsigned long long E = ~0ULL, L = 0ULL;
+                    #pragma unroll
+                    for (int b = 7; b >= 0; --b) {
+                        const unsigned long long B =
+                            __ballot((my_code >> b) & 1);
+                        if ((my_code >> b) & 1) {
+                            L |= E & ~B;
+                            E &= B;
+                        } else {
+                            E &= ~B;
+                        }
                     }
+                    const unsigned long long le_mask = (L | E) & mask;
+                    const int cnt = __popcll(le_mask);
+                    const int cnt1 = __popcll(le_mask & lab_mask);
                     double sc = -1.0e300;
                     int sb = 0x7FFFFFFF, snl = 0;
                     if (in && my_code < cmax) {

@@ -154,6 +154,10 @@ std::vector<at::Tensor> forest_fit(
     a.mid_cap = (int)mid_cap;
     int* st = state.data_ptr<int>();
     a.pool_count = st + 1;   // kernel indexes [wp * 4]
+    // wave-parallel mid-subtree mode: RF / ET (max_features <= 4); the
+    // env var forces the block-serial DFS for same-box A/B runs
+    a.wave_mid = (max_features <= WAVE_CANDS &&
+                  !getenv("FLAKE16_NO_WAVE_MID")) ? 1 : 0;
 
     int GRID = 4096;
     if (const char* e = getenv("FLAKE16_FIT_GRID")) GRID = atoi(e);
@@ -243,18 +247,77 @@ std::vector<at::Tensor> forest_predict_confusion(
 // ---------------------------------------------------------------------------
 // knn / balancing / binning
 // ---------------------------------------------------------------------------
+// Shared implementation: MFMA two-phase kernel by default (matrix cores
+// compute the fp32 distance filter; survivors are re-checked in exact
+// fp64 — output bits identical to the scalar kernels), scalar tile
+// kernels under FLAKE16_NO_MFMA_KNN=1 for A/B runs.
+static at::Tensor knn_run(at::Tensor X, at::Tensor seg_off, const int* so,
+                          int n_seg, int64_t k, bool skip_identity) {
+    const int R = X.size(0);
+    hipStream_t stream = current_stream();
+    auto out = at::empty({R, k}, X.options().dtype(at::kInt));
+
+    if (getenv("FLAKE16_NO_MFMA_KNN")) {
+        std::vector<int> blk(n_seg + 1);
+        blk[0] = 0;
+        for (int s = 0; s < n_seg; ++s) {
+            int n = so[s + 1] - so[s];
+            blk[s + 1] = blk[s] + (n + KNN_BLK - 1) / KNN_BLK;
+        }
+        auto seg_blk = at::from_blob(blk.data(), {n_seg + 1}, at::kInt)
+                           .to(X.device());
+        if (blk[n_seg] > 0)
+            knn_segmented_kernel<<<blk[n_seg], KNN_BLK, 0, stream>>>(
+                X.data_ptr<float>(), seg_off.data_ptr<int>(),
+                seg_blk.data_ptr<int>(), n_seg, (int)k,
+                skip_identity ? 1 : 0, out.data_ptr<int>());
+        return out;
+    }
+
+    std::vector<int> blk(n_seg + 1), segrow(R);
+    blk[0] = 0;
+    for (int s = 0; s < n_seg; ++s) {
+        int n = so[s + 1] - so[s];
+        blk[s + 1] = blk[s] + (n + 63) / 64;
+        for (int r = so[s]; r < so[s + 1]; ++r) segrow[r] = s;
+    }
+    auto seg_blk = at::from_blob(blk.data(), {n_seg + 1}, at::kInt)
+                       .to(X.device());
+    auto seg_of_row = at::from_blob(segrow.data(), {R}, at::kInt)
+                          .to(X.device());
+    auto norms = at::empty({R}, X.options());
+    auto segmax = at::zeros({n_seg}, X.options().dtype(at::kInt));
+    auto fb_list = at::empty({R}, X.options().dtype(at::kInt));
+    auto fb_count = at::zeros({1}, X.options().dtype(at::kInt));
+
+    knn_norms_kernel<<<(R + 255) / 256, 256, 0, stream>>>(
+        X.data_ptr<float>(), R, seg_of_row.data_ptr<int>(),
+        norms.data_ptr<float>(), segmax.data_ptr<int>());
+    if (blk[n_seg] > 0) {
+        knn_mfma_kernel<<<blk[n_seg], KNN_BLK, 0, stream>>>(
+            X.data_ptr<float>(), seg_off.data_ptr<int>(),
+            seg_blk.data_ptr<int>(), n_seg, (int)k, skip_identity ? 1 : 0,
+            norms.data_ptr<float>(), segmax.data_ptr<int>(),
+            out.data_ptr<int>(), fb_list.data_ptr<int>(),
+            fb_count.data_ptr<int>());
+        knn_fallback_kernel<<<(R + 255) / 256, 256, 0, stream>>>(
+            X.data_ptr<float>(), seg_off.data_ptr<int>(),
+            seg_of_row.data_ptr<int>(), (int)k, skip_identity ? 1 : 0,
+            fb_list.data_ptr<int>(), fb_count.data_ptr<int>(),
+            out.data_ptr<int>());
+    }
+    return out;
+}
+
 at::Tensor knn(at::Tensor X, int64_t k, bool skip_identity) {
     const at::cuda::OptionalCUDAGuard guard(X.device());
     TORCH_CHECK(X.is_cuda() && X.dtype() == at::kFloat &&
                 X.size(1) == FPAD && X.is_contiguous());
     TORCH_CHECK(k >= 1 && k <= KMAX);
     const int n = X.size(0);
-    auto out = at::empty({n, k}, X.options().dtype(at::kInt));
-    const int grid = (n + KNN_BLK - 1) / KNN_BLK;
-    knn_kernel<<<grid, KNN_BLK, 0, current_stream()>>>(
-        X.data_ptr<float>(), n, (int)k, skip_identity ? 1 : 0,
-        out.data_ptr<int>());
-    return out;
+    int so[2] = {0, n};
+    auto seg_off = at::from_blob(so, {2}, at::kInt).to(X.device());
+    return knn_run(X, seg_off, so, 1, k, skip_identity);
 }
 
 at::Tensor knn_segmented(at::Tensor X, at::Tensor seg_off, int64_t k,
@@ -264,26 +327,9 @@ at::Tensor knn_segmented(at::Tensor X, at::Tensor seg_off, int64_t k,
                 X.size(1) == FPAD && X.is_contiguous());
     TORCH_CHECK(k >= 1 && k <= KMAX);
     const int n_seg = seg_off.size(0) - 1;
-    const int R = X.size(0);
-
     auto seg_off_cpu = seg_off.to(at::kCPU);
-    const int* so = seg_off_cpu.data_ptr<int>();
-    std::vector<int> blk(n_seg + 1);
-    blk[0] = 0;
-    for (int s = 0; s < n_seg; ++s) {
-        int n = so[s + 1] - so[s];
-        blk[s + 1] = blk[s] + (n + KNN_BLK - 1) / KNN_BLK;
-    }
-    auto seg_blk = at::from_blob(blk.data(), {n_seg + 1}, at::kInt)
-                       .to(X.device());
-
-    auto out = at::empty({R, k}, X.options().dtype(at::kInt));
-    if (blk[n_seg] > 0)
-        knn_segmented_kernel<<<blk[n_seg], KNN_BLK, 0, current_stream()>>>(
-            X.data_ptr<float>(), seg_off.data_ptr<int>(),
-            seg_blk.data_ptr<int>(), n_seg, (int)k, skip_identity ? 1 : 0,
-            out.data_ptr<int>());
-    return out;
+    return knn_run(X, seg_off, seg_off_cpu.data_ptr<int>(), n_seg, k,
+                   skip_identity);
 }
 
 at::Tensor smote_interpolate(at::Tensor X, at::Tensor min_rows,

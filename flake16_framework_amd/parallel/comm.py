@@ -38,17 +38,35 @@ def rank_world():
 
 
 def init_from_env(backend=None):
-    """Initialize the process group from torchrun env vars if present."""
+    """Initialize the process group from torchrun env vars if present.
+
+    Backend selection: RCCL ("nccl") when there is one GPU per rank — the
+    production layout.  When ranks OVERSUBSCRIBE the visible GPUs (e.g. a
+    2-rank smoke test on a 1-GPU box) RCCL refuses duplicate devices
+    ("Duplicate GPU detected"), so collectives fall back to gloo on host
+    memory while the compute path stays on CUDA; payloads here are ~KB, so
+    the transport does not matter for validation runs."""
     if is_initialized() or "RANK" not in os.environ:
         return rank_world()
     import torch
+    world = int(os.environ.get("WORLD_SIZE", "1"))
     if backend is None:
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
-    if backend == "nccl":
+        if torch.cuda.is_available() and world <= torch.cuda.device_count():
+            backend = "nccl"
+        else:
+            backend = "gloo"
+    if torch.cuda.is_available():
         torch.cuda.set_device(min(int(os.environ.get("LOCAL_RANK", "0")),
                                   torch.cuda.device_count() - 1))
     dist().init_process_group(backend=backend)
     return rank_world()
+
+
+def collective_device():
+    """Device collective tensors should live on for the current backend."""
+    if is_initialized() and dist().get_backend() == "nccl":
+        return "cuda"
+    return "cpu"
 
 
 def shard_cells(world, rank, n_cells=None):
