@@ -185,12 +185,19 @@ typedef float v4f __attribute__((ext_vector_type(4)));
 
 #define KNN_POOL_M 12          // per-lane phase-1 list length (>= k + 4)
 
-// Per-row squared norms (fp32) + per-segment max via integer atomicMax
-// (IEEE order == integer order for non-negative floats).
+__device__ __forceinline__ int knn_seg_of_row(
+    const int* __restrict__ seg_off, int n_seg, int row) {
+    int lo = 0, hi = n_seg;
+    while (lo + 1 < hi) {
+        int mid = (lo + hi) >> 1;
+        if (seg_off[mid] <= row) lo = mid; else hi = mid;
+    }
+    return lo;
+}
+
+// Per-row squared norms (fp32).
 __global__ void knn_norms_kernel(const float* __restrict__ X, int R,
-                                 const int* __restrict__ seg_of_row,
-                                 float* __restrict__ norms,
-                                 int* __restrict__ segmax) {
+                                 float* __restrict__ norms) {
     const int r = blockIdx.x * blockDim.x + threadIdx.x;
     if (r >= R) return;
     float s = 0.0f;
@@ -200,7 +207,6 @@ __global__ void knn_norms_kernel(const float* __restrict__ X, int R,
         s = fmaf(v, v, s);
     }
     norms[r] = s;
-    atomicMax(&segmax[seg_of_row[r]], __float_as_int(s));
 }
 
 // One 256-thread block = 4 waves = 64 queries of one segment; each wave
@@ -211,7 +217,6 @@ __global__ void knn_mfma_kernel(const float* __restrict__ X,
                                 const int* __restrict__ seg_blk,
                                 int n_seg, int k, int skip_identity,
                                 const float* __restrict__ norms,
-                                const int* __restrict__ segmax,
                                 int* __restrict__ out,
                                 int* __restrict__ fb_list,
                                 int* __restrict__ fb_count) {
@@ -252,31 +257,47 @@ __global__ void knn_mfma_kernel(const float* __restrict__ X,
     #pragma unroll
     for (int j = 0; j < KNN_POOL_M; ++j) { td[j] = 1.0e30f; ti[j] = -1; }
     float discard_min = 1.0e30f;
+    float worst = 1.0e30f;       // == td[KNN_POOL_M - 1], kept scalar
+    const float qn = q_ok ? norms[base + myq] : 0.0f;
 
-    for (int c0 = 0; c0 < n; c0 += 16) {
-        // A fragments: candidate row jcol of this tile, same K-slices
-        const int ca = c0 + jcol;
-        float cf[4];
+    // two 16-candidate tiles per iteration: independent accumulator
+    // chains hide the MFMA dependent latency, loads amortize
+    for (int c0 = 0; c0 < n; c0 += 32) {
+        const int ca0 = c0 + jcol;
+        const int ca1 = c0 + 16 + jcol;
+        float cf0[4], cf1[4], nrm[8];
         #pragma unroll
-        for (int kk = 0; kk < 4; ++kk)
-            cf[kk] = ca < n
-                ? X[(size_t)(base + ca) * FPAD + 4 * kk + kfrag] : 0.0f;
-
-        v4f acc = {0.0f, 0.0f, 0.0f, 0.0f};
+        for (int kk = 0; kk < 4; ++kk) {
+            cf0[kk] = ca0 < n
+                ? X[(size_t)(base + ca0) * FPAD + 4 * kk + kfrag] : 0.0f;
+            cf1[kk] = ca1 < n
+                ? X[(size_t)(base + ca1) * FPAD + 4 * kk + kfrag] : 0.0f;
+        }
         #pragma unroll
-        for (int kk = 0; kk < 4; ++kk)
-            acc = __builtin_amdgcn_mfma_f32_16x16x4f32(cf[kk], qf[kk],
-                                                       acc, 0, 0, 0);
+        for (int t = 0; t < 8; ++t) {
+            const int c = c0 + (t >> 2) * 16 + kfrag * 4 + (t & 3);
+            nrm[t] = c < n ? norms[base + c] : 0.0f;
+        }
 
-        // D rows this lane owns: candidates c0 + kfrag*4 + reg
+        v4f acc0 = {0.0f, 0.0f, 0.0f, 0.0f};
+        v4f acc1 = {0.0f, 0.0f, 0.0f, 0.0f};
+        #pragma unroll
+        for (int kk = 0; kk < 4; ++kk) {
+            acc0 = __builtin_amdgcn_mfma_f32_16x16x4f32(cf0[kk], qf[kk],
+                                                        acc0, 0, 0, 0);
+            acc1 = __builtin_amdgcn_mfma_f32_16x16x4f32(cf1[kk], qf[kk],
+                                                        acc1, 0, 0, 0);
+        }
+
+        // D rows this lane owns: candidates c0 (+16) + kfrag*4 + reg
         if (q_ok) {
-            const float qn = norms[base + myq];
             #pragma unroll
-            for (int reg = 0; reg < 4; ++reg) {
-                const int c = c0 + kfrag * 4 + reg;
+            for (int t = 0; t < 8; ++t) {
+                const int c = c0 + (t >> 2) * 16 + kfrag * 4 + (t & 3);
                 if (c >= n || (skip_identity && c == myq)) continue;
-                const float d32 = qn + norms[base + c] - 2.0f * acc[reg];
-                if (d32 < td[KNN_POOL_M - 1]) {
+                const float g = (t < 4) ? acc0[t & 3] : acc1[t & 3];
+                const float d32 = qn + nrm[t] - 2.0f * g;
+                if (d32 < worst) {
                     // displaced entry becomes a discard
                     if (ti[KNN_POOL_M - 1] >= 0)
                         discard_min = fminf(discard_min,
@@ -289,6 +310,7 @@ __global__ void knn_mfma_kernel(const float* __restrict__ X,
                     }
                     td[j] = d32;
                     ti[j] = c;
+                    worst = td[KNN_POOL_M - 1];
                 } else {
                     discard_min = fminf(discard_min, d32);
                 }
@@ -331,9 +353,15 @@ __global__ void knn_mfma_kernel(const float* __restrict__ X,
                 }
             }
         }
-        const float slack = 4.0e-5f * __int_as_float(segmax[seg]);
+        // Per-query error bound: |d32 - d| <= ~24 eps32 B with
+        // B = (|q| + |c|)^2 <= (2|q| + sqrt(tau))^2 for contenders
+        // (24 roundings of magnitude <= B); slack carries an 8x margin.
+        const float qnm = norms[base + q];
+        const float rt = 2.0f * sqrtf(qnm) + sqrtf(fmaxf(topd[k - 1], 0.0f));
+        const float slack = 1.2e-5f * rt * rt + 1.0e-30f;
         if (dmin <= topd[k - 1] + slack) {
             // phase-1 list may have dropped a contender: exact fallback
+            atomicAdd(&fb_count[1 + seg], 1);
             fb_list[atomicAdd(fb_count, 1)] = base + q;
             return;
         }
@@ -385,51 +413,124 @@ __global__ void knn_mfma_kernel(const float* __restrict__ X,
     }
 }
 
-// Exact scalar scan for fallback queries (one thread per flagged query).
+// Fallback plumbing: flagged queries are regrouped by segment into
+// 256-padded regions so the exact re-scan runs with the same LDS
+// candidate tiling as the scalar kernel — the fallback then costs what
+// the scalar path cost, but only for the flagged fraction.
+
+// fb_count[0] = total flagged; fb_count[1 + s] = flagged in segment s.
+// One thread: 256-padded exclusive offsets + cursor reset.
+__global__ void knn_fb_scan_kernel(const int* __restrict__ fb_count,
+                                   int n_seg, int* __restrict__ fb_off,
+                                   int* __restrict__ fb_cursor) {
+    if (threadIdx.x != 0 || blockIdx.x != 0) return;
+    int acc = 0;
+    for (int s = 0; s < n_seg; ++s) {
+        fb_off[s] = acc;
+        fb_cursor[s] = 0;
+        acc += (fb_count[1 + s] + KNN_BLK - 1) / KNN_BLK * KNN_BLK;
+    }
+    fb_off[n_seg] = acc;
+}
+
+// -1-fill the used region of fb_sorted (device-bounded: costs nothing
+// when no query was flagged).
+__global__ void knn_fb_fill_kernel(const int* __restrict__ fb_off,
+                                   int n_seg, int* __restrict__ fb_sorted) {
+    const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < fb_off[n_seg]) fb_sorted[i] = -1;
+}
+
+// Scatter flagged rows into their segment's region (fb_sorted pre-filled
+// with -1; every KNN_BLK-aligned block of it is single-segment).
+__global__ void knn_fb_scatter_kernel(const int* __restrict__ fb_list,
+                                      const int* __restrict__ fb_count,
+                                      const int* __restrict__ seg_off,
+                                      int n_seg,
+                                      const int* __restrict__ fb_off,
+                                      int* __restrict__ fb_cursor,
+                                      int* __restrict__ fb_sorted) {
+    const int i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= fb_count[0]) return;
+    const int row = fb_list[i];
+    const int seg = knn_seg_of_row(seg_off, n_seg, row);
+    const int pos = atomicAdd(&fb_cursor[seg], 1);
+    fb_sorted[fb_off[seg] + pos] = row;
+}
+
+// Exact scan for the flagged queries, LDS-tiled over the segment's
+// candidates (numerics identical to knn_segmented_kernel).
+__launch_bounds__(KNN_BLK)
 __global__ void knn_fallback_kernel(const float* __restrict__ X,
                                     const int* __restrict__ seg_off,
-                                    const int* __restrict__ seg_of_row,
-                                    int k, int skip_identity,
-                                    const int* __restrict__ fb_list,
-                                    const int* __restrict__ fb_count,
+                                    int n_seg, int k, int skip_identity,
+                                    const int* __restrict__ fb_sorted,
+                                    const int* __restrict__ fb_off,
                                     int* __restrict__ out) {
-    const int i = blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= *fb_count) return;
-    const int row = fb_list[i];
-    const int seg = seg_of_row[row];
+    __shared__ float tile[KNN_BLK][FPAD];
+    __shared__ int sh_seg;
+
+    const long slot = (long)blockIdx.x * KNN_BLK + threadIdx.x;
+    if (threadIdx.x == 0) sh_seg = -1;
+    __syncthreads();
+    const int row = slot < fb_off[n_seg] ? fb_sorted[slot] : -1;
+    if (row >= 0)
+        sh_seg = knn_seg_of_row(seg_off, n_seg, row);  // benign same-value race
+    __syncthreads();
+    const int seg = sh_seg;
+    if (seg < 0) return;   // all-padding block
     const int base = seg_off[seg];
     const int n = seg_off[seg + 1] - base;
     const int q = row - base;
 
-    double qv[FPAD];
-    #pragma unroll
-    for (int f = 0; f < FPAD; ++f)
-        qv[f] = (double)X[(size_t)row * FPAD + f];
+    float qv[FPAD];
+    if (row >= 0) {
+        #pragma unroll
+        for (int f = 0; f < FPAD; ++f)
+            qv[f] = X[(size_t)row * FPAD + f];
+    }
 
     double bd[KMAX];
     int bi[KMAX];
     for (int j = 0; j < KMAX; ++j) { bd[j] = 1.0e300; bi[j] = -1; }
 
-    for (int c = 0; c < n; ++c) {
-        if (skip_identity && c == q) continue;
-        double d = 0.0;
-        #pragma unroll
-        for (int f = 0; f < FPAD; ++f) {
-            const double diff = qv[f] - (double)X[(size_t)(base + c) * FPAD + f];
-            d = d + diff * diff;
+    for (int tb = 0; tb < n; tb += KNN_BLK) {
+        const int c = tb + threadIdx.x;
+        if (c < n) {
+            #pragma unroll
+            for (int f = 0; f < FPAD; ++f)
+                tile[threadIdx.x][f] = X[(size_t)(base + c) * FPAD + f];
         }
-        if (d < bd[k - 1]) {
-            int j = k - 1;
-            while (j > 0 && d < bd[j - 1]) {
-                bd[j] = bd[j - 1];
-                bi[j] = bi[j - 1];
-                --j;
+        __syncthreads();
+
+        if (row >= 0) {
+            const int tn = min(KNN_BLK, n - tb);
+            for (int t = 0; t < tn; ++t) {
+                const int cand = tb + t;
+                if (skip_identity && cand == q) continue;
+                double d = 0.0;
+                #pragma unroll
+                for (int f = 0; f < FPAD; ++f) {
+                    const double diff = (double)qv[f] - (double)tile[t][f];
+                    d = d + diff * diff;
+                }
+                if (d < bd[k - 1]) {
+                    int j = k - 1;
+                    while (j > 0 && d < bd[j - 1]) {
+                        bd[j] = bd[j - 1];
+                        bi[j] = bi[j - 1];
+                        --j;
+                    }
+                    bd[j] = d;
+                    bi[j] = cand;
+                }
             }
-            bd[j] = d;
-            bi[j] = c;
         }
+        __syncthreads();
     }
-    for (int j = 0; j < k; ++j) out[(size_t)row * k + j] = bi[j];
+
+    if (row >= 0)
+        for (int j = 0; j < k; ++j) out[(size_t)row * k + j] = bi[j];
 }
 
 // smote_kernel: one thread per synthetic sample.

@@ -274,37 +274,53 @@ static at::Tensor knn_run(at::Tensor X, at::Tensor seg_off, const int* so,
         return out;
     }
 
-    std::vector<int> blk(n_seg + 1), segrow(R);
+    std::vector<int> blk(n_seg + 1);
     blk[0] = 0;
     for (int s = 0; s < n_seg; ++s) {
         int n = so[s + 1] - so[s];
         blk[s + 1] = blk[s] + (n + 63) / 64;
-        for (int r = so[s]; r < so[s + 1]; ++r) segrow[r] = s;
     }
     auto seg_blk = at::from_blob(blk.data(), {n_seg + 1}, at::kInt)
                        .to(X.device());
-    auto seg_of_row = at::from_blob(segrow.data(), {R}, at::kInt)
-                          .to(X.device());
     auto norms = at::empty({R}, X.options());
-    auto segmax = at::zeros({n_seg}, X.options().dtype(at::kInt));
     auto fb_list = at::empty({R}, X.options().dtype(at::kInt));
-    auto fb_count = at::zeros({1}, X.options().dtype(at::kInt));
+    auto fb_count = at::zeros({1 + n_seg}, X.options().dtype(at::kInt));
+    auto fb_off = at::empty({n_seg + 1}, X.options().dtype(at::kInt));
+    auto fb_cursor = at::empty({n_seg}, X.options().dtype(at::kInt));
+    const long fb_cap = (long)R + (long)KNN_BLK * n_seg;
+    auto fb_sorted = at::empty({fb_cap}, X.options().dtype(at::kInt));
 
     knn_norms_kernel<<<(R + 255) / 256, 256, 0, stream>>>(
-        X.data_ptr<float>(), R, seg_of_row.data_ptr<int>(),
-        norms.data_ptr<float>(), segmax.data_ptr<int>());
+        X.data_ptr<float>(), R, norms.data_ptr<float>());
     if (blk[n_seg] > 0) {
         knn_mfma_kernel<<<blk[n_seg], KNN_BLK, 0, stream>>>(
             X.data_ptr<float>(), seg_off.data_ptr<int>(),
             seg_blk.data_ptr<int>(), n_seg, (int)k, skip_identity ? 1 : 0,
-            norms.data_ptr<float>(), segmax.data_ptr<int>(),
-            out.data_ptr<int>(), fb_list.data_ptr<int>(),
-            fb_count.data_ptr<int>());
-        knn_fallback_kernel<<<(R + 255) / 256, 256, 0, stream>>>(
-            X.data_ptr<float>(), seg_off.data_ptr<int>(),
-            seg_of_row.data_ptr<int>(), (int)k, skip_identity ? 1 : 0,
+            norms.data_ptr<float>(), out.data_ptr<int>(),
+            fb_list.data_ptr<int>(), fb_count.data_ptr<int>());
+        // regroup flagged queries by segment (256-padded) and re-scan
+        // them exactly with LDS candidate tiling; all plumbing is
+        // device-bounded — when nothing was flagged these kernels are
+        // a few empty dispatches
+        knn_fb_scan_kernel<<<1, 1, 0, stream>>>(
+            fb_count.data_ptr<int>(), n_seg, fb_off.data_ptr<int>(),
+            fb_cursor.data_ptr<int>());
+        knn_fb_fill_kernel<<<(int)((fb_cap + 255) / 256), 256, 0, stream>>>(
+            fb_off.data_ptr<int>(), n_seg, fb_sorted.data_ptr<int>());
+        knn_fb_scatter_kernel<<<(R + 255) / 256, 256, 0, stream>>>(
             fb_list.data_ptr<int>(), fb_count.data_ptr<int>(),
-            out.data_ptr<int>());
+            seg_off.data_ptr<int>(), n_seg, fb_off.data_ptr<int>(),
+            fb_cursor.data_ptr<int>(), fb_sorted.data_ptr<int>());
+        knn_fallback_kernel<<<(int)((fb_cap + KNN_BLK - 1) / KNN_BLK),
+                              KNN_BLK, 0, stream>>>(
+            X.data_ptr<float>(), seg_off.data_ptr<int>(), n_seg, (int)k,
+            skip_identity ? 1 : 0, fb_sorted.data_ptr<int>(),
+            fb_off.data_ptr<int>(), out.data_ptr<int>());
+    }
+    if (getenv("FLAKE16_KNN_DEBUG")) {
+        const int fb = fb_count.narrow(0, 0, 1).to(at::kCPU).item<int>();
+        fprintf(stderr, "[knn] R=%d n_seg=%d k=%ld fallback=%d\n",
+                R, n_seg, (long)k, fb);
     }
     return out;
 }
