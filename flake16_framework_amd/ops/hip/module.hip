@@ -65,18 +65,42 @@ std::vector<at::Tensor> forest_fit(
     auto j_node_off = at::from_blob(node_off.data(), {J}, at::kLong)
                           .to(codes.device());
 
-    auto nfeat = at::full({Ntot}, LEAF_SENTINEL, opts_i32);
+    // Workspace sizes are rounded to a coarse granularity so cells of
+    // similar size hit the same caching-allocator buckets: per-cell
+    // workspaces here reach GBs at large N, and exact (per-balance-group)
+    // sizes would force hipMalloc/hipFree churn — whose device syncs
+    // serialize the concurrent cell streams (measured: the N=40k sweep
+    // ran at ~79% single-stream occupancy before this).
+    const long GRAN = 1L << 23;
+    const long S_alloc = (S + GRAN - 1) / GRAN * GRAN;
+    const long Ntot_alloc = (Ntot + GRAN - 1) / GRAN * GRAN;
+
+    auto nfeat = at::full({Ntot_alloc}, LEAF_SENTINEL, opts_i32);
     // only nfeat needs initialization (LEAF sentinel); the other node
     // fields are written for every reached node before any read
-    auto nsplit = at::empty({Ntot}, opts_i32);
-    auto nleft = at::empty({Ntot}, opts_i32);
-    auto ncnt0 = at::empty({Ntot}, opts_f32);
-    auto ncnt1 = at::empty({Ntot}, opts_f32);
+    auto nsplit = at::empty({Ntot_alloc}, opts_i32);
+    auto nleft = at::empty({Ntot_alloc}, opts_i32);
+    auto ncnt0 = at::empty({Ntot_alloc}, opts_f32);
+    auto ncnt1 = at::empty({Ntot_alloc}, opts_f32);
     auto node_alloc = at::zeros({J}, opts_i32);
 
-    auto sidx_a = at::empty({S}, opts_i32);
-    auto sidx_b = at::empty({S}, opts_i32);
-    const long work_cap = S + J + 2;
+    auto sidx_a = at::empty({S_alloc}, opts_i32);
+    auto sidx_b = at::empty({S_alloc}, opts_i32);
+
+    // Histogram-subtraction pools (see forest.hip): sized for the worst
+    // per-level allocation, 2 slots per splitting node >= HIST_SAVE_MIN.
+    int HIST_SAVE_MIN = 2048;
+    if (const char* e = getenv("FLAKE16_HIST_SAVE_MIN"))
+        HIST_SAVE_MIN = atoi(e);
+    // cap bounds pool memory at ~2 GB per parity (16 KiB per slot);
+    // exhaustion is correct but silently degrades to full accumulation
+    const long pool_cap =
+        std::min<long>(2 * (S_alloc / HIST_SAVE_MIN) + 8, 131072);
+
+    // Level work queue: only nodes > MID_N samples (disjoint: <= S/MID_N)
+    // plus <= MID_N children carrying a subtraction slot (<= 2 per pool
+    // slot) and the J roots; overflow trips err_flag loudly.
+    const long work_cap = S_alloc / MID_N + 2 * pool_cap + J + 64;
     auto work_a = at::empty({work_cap * (long)sizeof(WorkItem)},
                             codes.options().dtype(at::kByte));
     auto work_b = at::empty({work_cap * (long)sizeof(WorkItem)},
@@ -87,23 +111,16 @@ std::vector<at::Tensor> forest_fit(
     auto state = at::zeros({8}, opts_i32);
     auto err = at::zeros({1}, opts_i32);
 
-    auto small_q = at::empty({(S + 2) * (long)sizeof(WorkItem)},
+    // Small queue: disjoint <= SMALL_N subtrees; S/4 covers any
+    // non-adversarial level (err_flag guards the theoretical S bound).
+    const long small_cap = S_alloc / 4 + 1024;
+    auto small_q = at::empty({small_cap * (long)sizeof(WorkItem)},
                              codes.options().dtype(at::kByte));
 
     // mid-subtree queue: disjoint ranges of > SMALL_N samples each
-    const long mid_cap = S / (SMALL_N + 1) + 8;
+    const long mid_cap = S_alloc / (SMALL_N + 1) + 64;
     auto mid_q = at::empty({mid_cap * (long)sizeof(WorkItem)},
                            codes.options().dtype(at::kByte));
-
-    // Histogram-subtraction pools (see forest.hip): sized for the worst
-    // per-level allocation, 2 slots per splitting node >= HIST_SAVE_MIN.
-    int HIST_SAVE_MIN = 2048;
-    if (const char* e = getenv("FLAKE16_HIST_SAVE_MIN"))
-        HIST_SAVE_MIN = atoi(e);
-    // cap bounds pool memory at ~2 GB per parity (16 KiB per slot);
-    // exhaustion is correct but silently degrades to full accumulation
-    const long pool_cap =
-        std::min<long>(2 * (S / HIST_SAVE_MIN) + 8, 131072);
     auto hist_pool0 = at::empty({pool_cap * FPAD * 256},
                                 codes.options().dtype(at::kInt));
     auto hist_pool1 = at::empty({pool_cap * FPAD * 256},
@@ -149,7 +166,7 @@ std::vector<at::Tensor> forest_fit(
     a.pool_cap = (int)pool_cap;
     a.hist_save_min = HIST_SAVE_MIN;
     a.small = (WorkItem*)small_q.data_ptr();
-    a.small_cap = (int)(S + 2);
+    a.small_cap = (int)small_cap;
     a.mid = (WorkItem*)mid_q.data_ptr();
     a.mid_cap = (int)mid_cap;
     int* st = state.data_ptr<int>();
