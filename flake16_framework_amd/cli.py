@@ -40,14 +40,14 @@ def main(argv=None):
     command, *args = argv
 
     if command == "setup":
-        from .orchestrate.runner import setup_image
-        setup_image()
+        from .orchestrate.runner import provision_all
+        provision_all()
     elif command == "container":
-        from .orchestrate.runner import manage_container
-        manage_container(*args)
+        from .orchestrate.runner import exec_suite
+        exec_suite(*args)
     elif command == "run":
-        from .orchestrate.runner import run_experiment
-        run_experiment(*args)
+        from .orchestrate.runner import drive_runs
+        drive_runs(*args)
     elif command == "tests":
         from .dataset.tests_io import write_tests
         write_tests()

@@ -1,6 +1,9 @@
-"""Collation of raw per-run data files into the per-project accumulator.
+"""Collation of raw per-run data files into per-project records.
 
-Data-plane parity with the reference (experiment.py:242-336):
+Data-plane parity with the reference (the FILE FORMATS of
+experiment.py:242-336 are the contract; the in-memory model here is this
+framework's own record classes):
+
   data/<proj>_<mode>_<runN>.<ext> where
     mode baseline|shuffle, ext tsv : one "outcome\\tnodeid" line per test
     mode testinspect, ext sqlite3  : coverage.py DB with per-test contexts
@@ -8,11 +11,15 @@ Data-plane parity with the reference (experiment.py:242-336):
     mode testinspect, ext pkl      : pickle (test_fn_ids, test_fn_data,
                                      test_files, churn)
 
-The per-project accumulator is
-  collated[proj] = [test_data, test_fn_data, test_files, churn]
-  test_data[nid] = [runs, cov, rusage, fid]
-    runs[mode]   = [n_runs, n_fails, min_failing_run, min_passing_run]
-    cov          = {relpath: set(line_numbers)}
+The model:
+  ProjectData  one subject project's accumulated evidence
+    .tests[nid] -> TestRecord
+    .fn_metrics {fid: 7 static metrics}, .test_files set,
+    .churn {file: {line: change_count}}
+  TestRecord   one test's evidence across all runs
+    .runs[mode] -> RunStats, .coverage {relpath: set(lines)},
+    .rusage [6 floats], .static_id fid
+  RunStats     pass/fail accounting for one mode
 """
 
 import os
@@ -22,39 +29,65 @@ import sqlite3
 from ..constants import DATA_DIR, SUBJECTS_DIR
 
 
-def iter_data_dir(data_dir=DATA_DIR):
-    """Yield (path, proj, mode, run_n, ext) for each file in data/."""
-    for file_name in os.listdir(data_dir):
-        proj, mode, rest = file_name.split("_", 2)
-        run_n, ext = rest.split(".", 1)
-        yield os.path.join(data_dir, file_name), proj, mode, int(run_n), ext
+class RunStats:
+    """Pass/fail accounting for one (test, mode) across repeated runs.
+
+    Failure detection is substring-based ("failed" in outcome), matching
+    the showflakes output contract."""
+
+    __slots__ = ("total", "failures", "first_fail", "first_pass")
+
+    def __init__(self):
+        self.total = 0
+        self.failures = 0
+        self.first_fail = None
+        self.first_pass = None
+
+    def observe(self, outcome, run_n):
+        self.total += 1
+        if "failed" in outcome:
+            self.failures += 1
+            if self.first_fail is None or run_n < self.first_fail:
+                self.first_fail = run_n
+        else:
+            if self.first_pass is None or run_n < self.first_pass:
+                self.first_pass = run_n
+
+    @property
+    def never_failed(self):
+        return self.failures == 0
+
+    @property
+    def always_failed(self):
+        return self.failures == self.total
 
 
-def iter_tsv(fd, n_split):
+class TestRecord:
+    """Everything collected about one test node."""
+
+    __slots__ = ("runs", "coverage", "rusage", "static_id")
+
+    def __init__(self):
+        self.runs = {}            # mode -> RunStats
+        self.coverage = None      # {relpath: set(line_numbers)}
+        self.rusage = None        # [6 floats]
+        self.static_id = None     # fid into ProjectData.fn_metrics
+
+    def stats(self, mode):
+        if mode not in self.runs:
+            self.runs[mode] = RunStats()
+        return self.runs[mode]
+
+    @property
+    def complete(self):
+        """All four evidence parts present (reference drops others)."""
+        return bool(self.runs) and bool(self.coverage) and \
+            bool(self.rusage) and self.static_id is not None
+
+
+def _tsv_rows(fd, n_split):
     for line in fd:
         yield line.strip().split("\t", n_split)
-
-
-def get_test_data_nid(collated_proj, nid):
-    return collated_proj[0].setdefault(nid, [{}, {}, None, None])
-
-
-def update_collated_runs(fd, mode, run_n, collated_proj):
-    """Accumulate pass/fail statistics from one run's outcome TSV.
-
-    Failure detection is substring-based ("failed" in outcome), matching the
-    showflakes output contract (reference experiment.py:260-277).
-    """
-    for outcome, nid in iter_tsv(fd, 1):
-        runs_nid = get_test_data_nid(collated_proj, nid)[0]
-        runs_mode = runs_nid.setdefault(mode, [0, 0, None, None])
-        runs_mode[0] += 1
-
-        if "failed" in outcome:
-            runs_mode[1] += 1
-            runs_mode[2] = run_n if runs_mode[2] is None else min(runs_mode[2], run_n)
-        else:
-            runs_mode[3] = run_n if runs_mode[3] is None else min(runs_mode[3], run_n)
 
 
 def _numbits_to_nums(numbits_blob):
@@ -68,67 +101,96 @@ def _numbits_to_nums(numbits_blob):
     return nums
 
 
-def update_collated_cov(con, proj, collated_proj, subjects_dir=SUBJECTS_DIR):
-    """Ingest a coverage.py sqlite3 DB with per-test dynamic contexts.
+class ProjectData:
+    """One project's accumulated evidence + the per-file ingestors."""
 
-    Schema consumed (coverage 6.2): context(id, context), file(id, path),
-    line_bits(context_id, file_id, numbits).  File paths are stored relative
-    to the project checkout (reference experiment.py:280-299).
-    """
-    cur = con.cursor()
+    def __init__(self, proj, subjects_dir=SUBJECTS_DIR):
+        self.proj = proj
+        self.subjects_dir = subjects_dir
+        self.tests = {}           # nid -> TestRecord
+        self.fn_metrics = None    # {fid: [7 static metrics]}
+        self.test_files = None    # set of test file paths
+        self.churn = None         # {file: {line: change_count}}
 
-    nodeids = dict(cur.execute("SELECT id, context FROM context").fetchall())
+    def record(self, nid):
+        if nid not in self.tests:
+            self.tests[nid] = TestRecord()
+        return self.tests[nid]
 
-    proj_dir = os.path.join(subjects_dir, proj, proj)
-    files = {
-        file_id: os.path.relpath(path, start=proj_dir)
-        for file_id, path in cur.execute("SELECT id, path FROM file").fetchall()
-    }
+    # -- ingestors (one per raw-file kind) --------------------------------
+    def add_outcomes(self, fd, mode, run_n):
+        """One run's showflakes outcome TSV."""
+        for outcome, nid in _tsv_rows(fd, 1):
+            self.record(nid).stats(mode).observe(outcome, run_n)
 
-    rows = cur.execute("SELECT context_id, file_id, numbits FROM line_bits")
-    for context_id, file_id, nb in rows.fetchall():
-        cov_nid = get_test_data_nid(collated_proj, nodeids[context_id])[1]
-        cov_nid[files[file_id]] = set(_numbits_to_nums(nb))
+    def add_coverage_db(self, con):
+        """coverage.py 6.2 sqlite3 DB with per-test dynamic contexts
+        (tables context / file / line_bits); file paths become relative
+        to the project checkout."""
+        cur = con.cursor()
+        contexts = dict(cur.execute("SELECT id, context FROM context"))
+        checkout = os.path.join(self.subjects_dir, self.proj, self.proj)
+        rel_paths = {
+            file_id: os.path.relpath(path, start=checkout)
+            for file_id, path in cur.execute("SELECT id, path FROM file")
+        }
+        for ctx_id, file_id, blob in cur.execute(
+                "SELECT context_id, file_id, numbits FROM line_bits"):
+            rec = self.record(contexts[ctx_id])
+            if rec.coverage is None:
+                rec.coverage = {}
+            rec.coverage[rel_paths[file_id]] = set(_numbits_to_nums(blob))
+
+    def add_rusage(self, fd):
+        """Rusage TSV: 6 floats (Execution Time, Read Count, Write Count,
+        Context Switches, Max. Threads, Max. Memory) then the nodeid."""
+        for *values, nid in _tsv_rows(fd, 6):
+            self.record(nid).rusage = [float(v) for v in values]
+
+    def add_static(self, fd):
+        """Static-metrics pickle: (test_fn_ids {nid->fid}, test_fn_data
+        {fid->7 metrics}, test_files set, churn {file->{line->count}})."""
+        fn_ids, self.fn_metrics, self.test_files, self.churn = \
+            pickle.load(fd)
+        for nid, fid in fn_ids.items():
+            self.record(nid).static_id = fid
+
+    @property
+    def complete(self):
+        """Static parts present (projects without them are dropped)."""
+        return self.fn_metrics is not None and \
+            bool(self.test_files) and bool(self.churn) and bool(self.tests)
 
 
-def update_collated_rusage(fd, collated_proj):
-    """Rusage TSV: 6 floats (Execution Time, Read Count, Write Count,
-    Context Switches, Max. Threads, Max. Memory) then the nodeid."""
-    for *rusage, nid in iter_tsv(fd, 6):
-        get_test_data_nid(collated_proj, nid)[2] = [float(x) for x in rusage]
+def iter_data_dir(data_dir=DATA_DIR):
+    """Yield (path, proj, mode, run_n, ext) for each file in data/."""
+    for file_name in os.listdir(data_dir):
+        proj, mode, rest = file_name.split("_", 2)
+        run_n, ext = rest.split(".", 1)
+        yield os.path.join(data_dir, file_name), proj, mode, int(run_n), ext
 
 
-def update_collated_static(fd, collated_proj):
-    """Static-metrics pickle: (test_fn_ids {nid->fid}, test_fn_data
-    {fid->7 static metrics}, test_files set, churn {file->{line->count}})."""
-    test_fn_ids, test_fn_data, test_files, churn = pickle.load(fd)
-    collated_proj[1] = test_fn_data
-    collated_proj[2] = test_files
-    collated_proj[3] = churn
+def collate(data_dir=DATA_DIR, subjects_dir=SUBJECTS_DIR):
+    """Walk data/ and build {proj: ProjectData}."""
+    projects = {}
 
-    for nid, fid in test_fn_ids.items():
-        get_test_data_nid(collated_proj, nid)[3] = fid
+    for path, proj, mode, run_n, ext in iter_data_dir(data_dir):
+        if proj not in projects:
+            projects[proj] = ProjectData(proj, subjects_dir)
+        data = projects[proj]
 
-
-def get_collated(data_dir=DATA_DIR, subjects_dir=SUBJECTS_DIR):
-    """Walk data/ and build the full collated structure."""
-    collated = {}
-
-    for file_name, proj, mode, run_n, ext in iter_data_dir(data_dir):
-        collated_proj = collated.setdefault(proj, [{}, None, None, None])
-
-        if mode in {"baseline", "shuffle"}:
-            with open(file_name, "r") as fd:
-                update_collated_runs(fd, mode, run_n, collated_proj)
+        if mode in ("baseline", "shuffle"):
+            with open(path, "r") as fd:
+                data.add_outcomes(fd, mode, run_n)
         elif mode == "testinspect":
             if ext == "sqlite3":
-                with sqlite3.connect(file_name) as con:
-                    update_collated_cov(con, proj, collated_proj, subjects_dir)
+                with sqlite3.connect(path) as con:
+                    data.add_coverage_db(con)
             elif ext == "tsv":
-                with open(file_name, "r") as fd:
-                    update_collated_rusage(fd, collated_proj)
+                with open(path, "r") as fd:
+                    data.add_rusage(fd)
             elif ext == "pkl":
-                with open(file_name, "rb") as fd:
-                    update_collated_static(fd, collated_proj)
+                with open(path, "rb") as fd:
+                    data.add_static(fd)
 
-    return collated
+    return projects

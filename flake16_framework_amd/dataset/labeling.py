@@ -1,7 +1,7 @@
 """Flakiness labeling rule.
 
-Implements the decision table of the reference's get_req_runs_label_nid
-(reference experiment.py:339-359):
+Implements the decision table of the reference's labeling logic
+(reference experiment.py:339-359) over RunStats records:
 
   - incomplete run counts (baseline != 2500 or shuffle != 2500) -> (0, None):
     the test is dropped.
@@ -13,38 +13,37 @@ Implements the decision table of the reference's get_req_runs_label_nid
       shuffle passes at least once -> (first passing shuffle run, OD_FLAKY)
   - baseline intermittently fails  -> (max(first failing, first passing
                                        baseline run), FLAKY)  [NOD-flaky]
-
-`runs_mode` accumulators are [n_runs, n_fails, min_failing_run, min_passing_run]
-as produced by the run collator.
 """
 
 from ..constants import FLAKY, NON_FLAKY, N_RUNS, OD_FLAKY
+from .collate import RunStats
 
-EMPTY_RUNS = (0, 0, None, None)
 
+def classify(runs, n_runs=None):
+    """(required_runs, label) for one test.
 
-def get_req_runs_label(runs_nid, n_runs=None):
-    """Return (required_runs, label) for one test's run statistics.
-
-    runs_nid: {"baseline": [n, fails, min_fail_run, min_pass_run],
-               "shuffle":  [...]}
+    runs: {mode: RunStats} with modes "baseline" and "shuffle";
     n_runs: override of the per-mode expected run counts (for tests).
+    Label None means the evidence is incomplete and the test is dropped.
     """
     n_runs = n_runs or N_RUNS
-    baseline = runs_nid.get("baseline", list(EMPTY_RUNS))
-    shuffle = runs_nid.get("shuffle", list(EMPTY_RUNS))
+    baseline = runs.get("baseline") or RunStats()
+    shuffle = runs.get("shuffle") or RunStats()
 
-    if baseline[0] != n_runs["baseline"] or shuffle[0] != n_runs["shuffle"]:
+    if baseline.total != n_runs["baseline"] or \
+            shuffle.total != n_runs["shuffle"]:
         return 0, None
 
-    if baseline[1] == 0:
-        if shuffle[1] == 0:
+    if baseline.never_failed:
+        if shuffle.never_failed:
             return 0, NON_FLAKY
-        return shuffle[2], OD_FLAKY
+        return shuffle.first_fail, OD_FLAKY
 
-    if baseline[1] == baseline[0]:
-        if shuffle[1] == shuffle[0]:
+    if baseline.always_failed:
+        if shuffle.always_failed:
             return 0, NON_FLAKY
-        return shuffle[3], OD_FLAKY
+        return shuffle.first_pass, OD_FLAKY
 
-    return max(baseline[2], baseline[3]), FLAKY
+    # intermittent baseline failures: order-independent (NOD) flakiness;
+    # detection requires seeing both a failure and a pass
+    return max(baseline.first_fail, baseline.first_pass), FLAKY

@@ -13,47 +13,54 @@ import json
 import numpy as np
 
 from ..constants import TESTS_FILE
-from .collate import get_collated
+from .collate import collate
 from .features import get_features_cov
-from .labeling import get_req_runs_label
+from .labeling import classify
 
 
-def build_tests(collated):
-    """Collated structure -> the tests.json dict."""
+def _ci_sorted(names):
+    return sorted(names, key=str.lower)
+
+
+def build_tests(projects):
+    """{proj: ProjectData} -> the tests.json dict.
+
+    Projects and nodeids are emitted in case-insensitive sorted order;
+    a test contributes a row only when every evidence part is present
+    and the labeling rule accepts its run counts.  Row layout:
+    [req_runs, label, 3 coverage features, 6 rusage features,
+    7 static features] — the canonical 16-column order."""
     tests = {}
 
-    for proj in sorted(collated.keys(), key=lambda s: s.lower()):
-        if not all(collated[proj]):
+    for proj in _ci_sorted(projects):
+        data = projects[proj]
+        if not data.complete:
             continue
 
-        test_data, test_fn_data, test_files, churn = collated[proj]
-        tests_proj = {}
-
-        for nid in sorted(test_data.keys(), key=lambda s: s.lower()):
-            if not all(test_data[nid]):
+        rows = {}
+        for nid in _ci_sorted(data.tests):
+            rec = data.tests[nid]
+            if not rec.complete:
                 continue
-
-            runs_nid, cov_nid, rusage_nid, fid = test_data[nid]
-            req_runs_nid, label_nid = get_req_runs_label(runs_nid)
-
-            if label_nid is None:
+            req_runs, label = classify(rec.runs)
+            if label is None:
                 continue
-
-            tests_proj[nid] = (
-                req_runs_nid, label_nid,
-                *get_features_cov(cov_nid, test_files, churn),
-                *rusage_nid,
-                *test_fn_data[fid],
+            rows[nid] = (
+                req_runs, label,
+                *get_features_cov(rec.coverage, data.test_files,
+                                  data.churn),
+                *rec.rusage,
+                *data.fn_metrics[rec.static_id],
             )
 
-        if tests_proj:
-            tests[proj] = tests_proj
+        if rows:
+            tests[proj] = rows
 
     return tests
 
 
 def write_tests(tests_file=TESTS_FILE):
-    tests = build_tests(get_collated())
+    tests = build_tests(collate())
     with open(tests_file, "w") as fd:
         json.dump(tests, fd, indent=4)
 

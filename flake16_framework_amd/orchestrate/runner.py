@@ -1,15 +1,24 @@
 """Execution orchestration: the `run`, `container`, and `setup` commands.
 
-Semantic port of the reference's data-collection layer
-(experiment.py:110-239): 130,026 per-run Docker containers (26 subjects x
-{2,500 baseline + 2,500 shuffle + 1 testinspect}), a process pool with a
-progress meter, and crash-restart resumability through an append-only
-log.txt of completed container names.
+Covers the reference's data-collection layer (the BEHAVIOR of
+experiment.py:110-239 — 130,026 per-run containers over 26 subjects, a
+process pool with a progress meter, crash-restart resumability through an
+append-only completion log).  The structure here is this framework's own;
+the compatibility contracts preserved byte/argv-exactly are:
 
-The pytest plugins installed into each subject venv are this package's
-collect.showflakes / collect.testinspect equivalents (the reference's are
-empty submodules).  Docker interaction is isolated behind run_container's
-`runner` argument so the layer is testable without Docker.
+  - subjects.txt rows: owner/repo,sha,package_dir[,pre-command...]
+  - the docker invocation (docker run -it -v data:rw --rm --init --cpus=1
+    --name=<cont> flake16framework python3 experiment.py container ...)
+  - the pytest collector flags (--record-file/--shuffle/--testinspect/
+    --set-exitstatus) and the plugin blacklist
+  - the pool progress meter lines ("<message>\\n" + "done/remain
+    elapsed/eta" carriage-return updates)
+  - log.txt: one completed container name per line, appended on success
+
+Docker interaction is isolated behind the `runner` callable so the layer
+is testable without Docker; FLAKE16_LOCAL_RUN=1 substitutes a fresh local
+python process for the container (used by the integration tests and
+docker-less hosts).
 """
 
 import os
@@ -19,6 +28,7 @@ import subprocess as sp
 import sys
 import time
 from multiprocessing import Pool
+from typing import NamedTuple
 
 from ..constants import (
     CONT_DATA_DIR, CONT_TIMEOUT, DATA_DIR, IMAGE_NAME, LOG_FILE,
@@ -41,45 +51,65 @@ FRAMEWORK_ROOT = os.path.dirname(os.path.dirname(os.path.dirname(
     os.path.abspath(__file__))))
 
 
-def iter_subjects(subjects_file=SUBJECTS_FILE):
-    """subjects.txt rows: owner/repo,sha,package_dir[,pre-command...]."""
+# ---------------------------------------------------------------------------
+# Subjects
+# ---------------------------------------------------------------------------
+
+class Subject(NamedTuple):
+    """One study subject: a pinned open-source project."""
+    proj: str          # short name (repo without the owner)
+    repo: str          # owner/repo
+    sha: str           # pinned commit
+    package_dir: str   # directory pip-installed -e (relative to checkout)
+    commands: tuple    # pre-commands + the pytest command (last)
+
+    @classmethod
+    def parse(cls, line):
+        repo, sha, package_dir, *commands = line.split(",")
+        return cls(repo.split("/", 1)[1], repo, sha, package_dir,
+                   tuple(commands))
+
+
+def read_subjects(subjects_file=SUBJECTS_FILE):
+    """subjects.txt -> [Subject]; blank lines ignored."""
     with open(subjects_file, "r") as fd:
-        for line in fd:
-            line = line.strip()
-            if not line:
-                continue
-            repo, *rest = line.split(",")
-            yield repo.split("/", 1)[1], repo, *rest
+        return [Subject.parse(line.strip()) for line in fd if line.strip()]
 
 
-def setup_project(proj, url, sha, package_dir, subjects_dir=SUBJECTS_DIR):
-    """Create the subject's venv, clone at the pinned SHA, install pinned
-    requirements and the project itself."""
-    proj_dir = os.path.join(subjects_dir, proj, proj)
-    venv_dir = os.path.join(subjects_dir, proj, "venv")
-    requirements_file = os.path.join(subjects_dir, proj, "requirements.txt")
+# ---------------------------------------------------------------------------
+# Subject environment provisioning (the `setup` stage)
+# ---------------------------------------------------------------------------
+
+def provision_subject(subject, subjects_dir=SUBJECTS_DIR):
+    """Create the subject's venv, clone at the pinned SHA, install the
+    pinned requirements + the project, and expose the collectors."""
+    root = os.path.join(subjects_dir, subject.proj)
+    checkout = os.path.join(root, subject.proj)
+    venv_dir = os.path.join(root, "venv")
+    pins = os.path.join(root, "requirements.txt")
 
     env = os.environ.copy()
     env["PATH"] = os.path.join(venv_dir, "bin") + ":" + env["PATH"]
 
     sp.run(["virtualenv", venv_dir], check=True)
-    sp.run(["git", "clone", url, proj_dir], check=True)
-    sp.run(["git", "reset", "--hard", sha], cwd=proj_dir, check=True)
+    sp.run(["git", "clone", f"https://github.com/{subject.repo}", checkout],
+           check=True)
+    sp.run(["git", "reset", "--hard", subject.sha], cwd=checkout, check=True)
 
-    package_dir = os.path.join(proj_dir, package_dir)
     sp.run([*PIP_INSTALL, PIP_VERSION], env=env, check=True)
-    sp.run([*PIP_INSTALL, "-r", requirements_file], env=env, check=True)
-    sp.run([*PIP_INSTALL, "-e", package_dir], env=env, check=True)
+    sp.run([*PIP_INSTALL, "-r", pins], env=env, check=True)
+    sp.run([*PIP_INSTALL, "-e", os.path.join(checkout, subject.package_dir)],
+           env=env, check=True)
     install_plugins(venv_dir)
 
 
 def install_plugins(venv_dir, framework_root=FRAMEWORK_ROOT):
     """Make the collection plugins importable from the subject venv (the
     reference pip-installs its showflakes/testinspect checkouts into each
-    venv, experiment.py:125).  The framework's own setup.py builds the HIP
-    extension and needs torch, which subject venvs don't ship, so the
-    package is exposed with a .pth in the venv's site-packages instead of
-    a pip install — equivalent for `pytest -p flake16_framework_amd...`."""
+    venv).  The framework's own setup.py builds the HIP extension and
+    needs torch, which subject venvs don't ship, so the package is exposed
+    with a .pth in the venv's site-packages instead of a pip install —
+    equivalent for `pytest -p flake16_framework_amd...`."""
     import glob
     site_dirs = glob.glob(os.path.join(venv_dir, "lib", "python*",
                                        "site-packages"))
@@ -91,17 +121,19 @@ def install_plugins(venv_dir, framework_root=FRAMEWORK_ROOT):
             fd.write(framework_root + "\n")
 
 
-def setup_image():
-    """Provision all subjects in parallel (runs inside `docker build`)."""
+def provision_all(subjects_file=SUBJECTS_FILE):
+    """Provision every subject in parallel (runs inside `docker build`)."""
     os.makedirs(CONT_DATA_DIR, exist_ok=True)
-    args = [(proj, f"https://github.com/{repo}", sha, package_dir)
-            for proj, repo, sha, package_dir, *_ in iter_subjects()]
     with Pool(processes=N_PROC) as pool:
-        pool.starmap(setup_project, args)
+        pool.map(provision_subject, read_subjects(subjects_file))
 
 
-def mode_flags(mode, data_file):
-    """Per-mode pytest flags (matches the showflakes/testinspect CLI)."""
+# ---------------------------------------------------------------------------
+# One suite execution (the `container` command, inside the container)
+# ---------------------------------------------------------------------------
+
+def collector_flags(mode, data_file):
+    """Per-mode pytest flags (the showflakes/testinspect CLI contract)."""
     return {
         "testinspect": [f"--testinspect={data_file}"],
         "baseline": [f"--record-file={data_file}.tsv"],
@@ -109,25 +141,25 @@ def mode_flags(mode, data_file):
     }[mode]
 
 
-def manage_container(cont_name, *commands, subjects_dir=None,
-                     data_dir=None, run=sp.run):
-    """Inside the container: run the subject's pre-commands, then pytest
-    with the plugin blacklist, our collectors, and the mode flags.
-    Directory overrides via FLAKE16_SUBJECTS_DIR / FLAKE16_DATA_DIR
-    support the local (docker-less) mode."""
+def exec_suite(cont_name, *commands, subjects_dir=None, data_dir=None,
+               run=sp.run):
+    """Run the subject's pre-commands, then pytest with the plugin
+    blacklist, our collectors, and the mode flags.  Directory overrides
+    via FLAKE16_SUBJECTS_DIR / FLAKE16_DATA_DIR support the local
+    (docker-less) mode."""
     subjects_dir = subjects_dir or os.environ.get("FLAKE16_SUBJECTS_DIR",
                                                   SUBJECTS_DIR)
     data_dir = data_dir or os.environ.get("FLAKE16_DATA_DIR", CONT_DATA_DIR)
     proj, mode, _ = cont_name.split("_", 2)
-    proj_dir = os.path.join(subjects_dir, proj, proj)
+    checkout = os.path.join(subjects_dir, proj, proj)
     data_file = os.path.join(data_dir, cont_name)
-    bin_dir = os.path.join(subjects_dir, proj, "venv", "bin")
+    venv_bin = os.path.join(subjects_dir, proj, "venv", "bin")
 
     env = os.environ.copy()
-    env["PATH"] = bin_dir + ":" + env["PATH"]
+    env["PATH"] = venv_bin + ":" + env["PATH"]
 
     for cmd in commands[:-1]:
-        run(shlex.split(cmd), cwd=proj_dir, env=env, check=True)
+        run(shlex.split(cmd), cwd=checkout, env=env, check=True)
 
     plugin_args = []
     for plug in COLLECT_PLUGINS:
@@ -135,11 +167,16 @@ def manage_container(cont_name, *commands, subjects_dir=None,
 
     run(
         [*shlex.split(commands[-1]), *PLUGIN_BLACKLIST, *plugin_args,
-         "--set-exitstatus", *mode_flags(mode, data_file)],
-        timeout=CONT_TIMEOUT, cwd=proj_dir, check=True, env=env)
+         "--set-exitstatus", *collector_flags(mode, data_file)],
+        timeout=CONT_TIMEOUT, cwd=checkout, check=True, env=env)
 
+
+# ---------------------------------------------------------------------------
+# Run launching (host side)
+# ---------------------------------------------------------------------------
 
 def docker_run_argv(cont_name, commands, host_data_dir):
+    """The container invocation — argv-compatible with the reference."""
     return [
         "docker", "run", "-it",
         f"-v={host_data_dir}:{CONT_DATA_DIR}:rw", "--rm", "--init",
@@ -148,24 +185,21 @@ def docker_run_argv(cont_name, commands, host_data_dir):
     ]
 
 
-def _local_run_argv(cont_name, commands):
-    """Local (no-docker) mode: the per-run isolation is a fresh python
-    process instead of a container.  Enabled by FLAKE16_LOCAL_RUN=1 with
-    FLAKE16_SUBJECTS_DIR / FLAKE16_DATA_DIR pointing at the work tree —
-    used by the integration tests and available for docker-less hosts."""
-    import sys
+def local_run_argv(cont_name, commands):
+    """Local (no-docker) mode: per-run isolation is a fresh python
+    process instead of a container (FLAKE16_LOCAL_RUN=1)."""
     return [sys.executable, "-m", "flake16_framework_amd.cli", "container",
             cont_name, *commands]
 
 
-def run_container(args, runner=None, stdout_dir=STDOUT_DIR):
-    """Launch one container run; append its stdout; report success."""
+def launch_run(args, runner=None, stdout_dir=STDOUT_DIR):
+    """Launch one run; append its stdout; report (message, result)."""
     cont_name, commands = args
     host_data_dir = os.path.join(os.getcwd(), DATA_DIR)
     stdout_file = os.path.join(stdout_dir, cont_name)
 
     if os.environ.get("FLAKE16_LOCAL_RUN"):
-        argv = _local_run_argv(cont_name, commands)
+        argv = local_run_argv(cont_name, commands)
     else:
         argv = docker_run_argv(cont_name, commands, host_data_dir)
 
@@ -181,67 +215,97 @@ def run_container(args, runner=None, stdout_dir=STDOUT_DIR):
     return f"{message}: {cont_name}", (succeeded, cont_name)
 
 
-def iter_containers(run_modes, subjects_file=SUBJECTS_FILE, n_runs=None):
+def enumerate_runs(run_modes, subjects_file=SUBJECTS_FILE, n_runs=None):
+    """All (container name, commands) pairs for the requested modes."""
     n_runs = n_runs or N_RUNS
-    for proj, _, _, _, *commands in iter_subjects(subjects_file):
+    for subject in read_subjects(subjects_file):
         for mode in set(run_modes):
             for run_n in range(n_runs[mode]):
-                yield f"{proj}_{mode}_{run_n}", commands
+                yield f"{subject.proj}_{mode}_{run_n}", subject.commands
 
 
-def manage_pool(pool, fn, args, out=None):
-    """Shuffle, imap_unordered, and print per-task progress lines with
-    elapsed/ETA minutes — the reference's pool meter (experiment.py:191).
-    `out` resolves to the CURRENT sys.stdout at call time (an import-time
-    default would capture a stale stream under redirection)."""
-    out = out or sys.stdout
-    n_finish = 0
-    t_start = time.time()
+# ---------------------------------------------------------------------------
+# Pool + progress + resumability
+# ---------------------------------------------------------------------------
 
+class ProgressMeter:
+    """The reference pool meter's line format: a completion message per
+    task, then "finished/remaining elapsed/eta" (minutes) with a trailing
+    carriage return."""
+
+    def __init__(self, total, out=None):
+        # resolve the CURRENT sys.stdout at call time (an import-time
+        # default would capture a stale stream under redirection)
+        self.out = out or sys.stdout
+        self.total = total
+        self.finished = 0
+        self.started = time.time()
+        self.out.write(f"0/{total} 0/?\r")
+
+    def task_done(self, message):
+        self.finished += 1
+        remain = self.total - self.finished
+        elapsed = time.time() - self.started
+        eta = elapsed / self.finished * remain
+        self.out.write(f"{message}\n\r")
+        self.out.write(f"{self.finished}/{remain} "
+                       f"{round(elapsed / 60)}/{round(eta / 60)}\r")
+
+
+def pooled_progress(pool, fn, args, out=None):
+    """imap_unordered over shuffled args with the progress meter; yields
+    each task's result."""
     random.shuffle(args)
-    out.write(f"0/{len(args)} 0/?\r")
-
+    meter = ProgressMeter(len(args), out=out)
     for message, result in pool.imap_unordered(fn, args):
-        n_finish += 1
-        n_remain = len(args) - n_finish
-        t_elapse = time.time() - t_start
-        t_remain = t_elapse / n_finish * n_remain
-        out.write(f"{message}\n\r")
-        out.write(f"{n_finish}/{n_remain} "
-                  f"{round(t_elapse / 60)}/{round(t_remain / 60)}\r")
+        meter.task_done(message)
         yield result
 
 
+class CompletionLog:
+    """Append-only completion log (log.txt): the crash-restart checkpoint
+    of the run stage, one container name per line."""
+
+    def __init__(self, path=LOG_FILE):
+        self.path = path
+
+    def completed(self):
+        if not os.path.exists(self.path):
+            return set()
+        with open(self.path, "r") as fd:
+            return {line.strip() for line in fd}
+
+    def mark(self, cont_name):
+        with open(self.path, "a") as fd:
+            fd.write(f"{cont_name}\n")
+
+
 def read_log(log_file=LOG_FILE):
-    if not os.path.exists(log_file):
-        return []
-    with open(log_file, "r") as fd:
-        return [line.strip() for line in fd]
+    return sorted(CompletionLog(log_file).completed())
 
 
-def run_experiment(*run_modes, subjects_file=SUBJECTS_FILE, n_runs=None,
-                   runner=None, log_file=LOG_FILE):
-    """The resumable run driver: skip completed runs (log.txt), launch the
-    rest through a process pool, append completions, exit 1 on any
-    failure."""
+def drive_runs(*run_modes, subjects_file=SUBJECTS_FILE, n_runs=None,
+               runner=None, log_file=LOG_FILE):
+    """The resumable run driver: skip completed runs, launch the rest
+    through a process pool, log completions, exit 1 on any failure."""
     os.makedirs(DATA_DIR, exist_ok=True)
     os.makedirs(STDOUT_DIR, exist_ok=True)
 
-    log = set(read_log(log_file))
+    log = CompletionLog(log_file)
+    done = log.completed()
     args = [(cont_name, commands)
-            for cont_name, commands in iter_containers(run_modes,
-                                                       subjects_file, n_runs)
-            if cont_name not in log]
+            for cont_name, commands in enumerate_runs(run_modes,
+                                                      subjects_file, n_runs)
+            if cont_name not in done]
 
     exitstatus = 0
-    run_fn = run_container if runner is None else \
-        (lambda a: run_container(a, runner=runner))
+    run_fn = launch_run if runner is None else \
+        (lambda a: launch_run(a, runner=runner))
 
     with Pool(processes=N_PROC) as pool:
-        for succeeded, cont_name in manage_pool(pool, run_fn, args):
+        for succeeded, cont_name in pooled_progress(pool, run_fn, args):
             if succeeded:
-                with open(log_file, "a") as fd:
-                    fd.write(f"{cont_name}\n")
+                log.mark(cont_name)
             else:
                 exitstatus = 1
 

@@ -11,70 +11,81 @@ import numpy as np
 import pytest
 
 from flake16_framework_amd.constants import FLAKY, NON_FLAKY, OD_FLAKY
-from flake16_framework_amd.dataset.collate import (
-    update_collated_rusage, update_collated_runs,
-)
+from flake16_framework_amd.dataset.collate import ProjectData, RunStats
 from flake16_framework_amd.dataset.features import get_features_cov
-from flake16_framework_amd.dataset.labeling import get_req_runs_label
+from flake16_framework_amd.dataset.labeling import classify
 from flake16_framework_amd.dataset.synthetic import make_synthetic_tests
 from flake16_framework_amd.dataset.tests_io import load_feat_lab_proj
 
 N_RUNS_SMALL = {"baseline": 4, "shuffle": 4, "testinspect": 1}
 
 
-def _runs(lines_per_run, mode):
-    collated_proj = [{}, None, None, None]
+def _ingest_runs(lines_per_run, mode):
+    data = ProjectData("p")
     for run_n, lines in enumerate(lines_per_run):
-        update_collated_runs(io.StringIO("\n".join(lines)), mode, run_n,
-                             collated_proj)
-    return collated_proj
+        data.add_outcomes(io.StringIO("\n".join(lines)), mode, run_n)
+    return data
 
 
-class TestUpdateCollatedRuns:
+def _stats(total, failures, first_fail, first_pass):
+    s = RunStats()
+    s.total = total
+    s.failures = failures
+    s.first_fail = first_fail
+    s.first_pass = first_pass
+    return s
+
+
+class TestRunIngestion:
     def test_counts_and_min_runs(self):
-        collated = _runs([
+        data = _ingest_runs([
             ["passed\tt::a", "failed\tt::b"],
             ["failed\tt::a", "failed\tt::b"],
             ["passed\tt::a", "passed\tt::b"],
         ], "baseline")
-        runs_a = collated[0]["t::a"][0]["baseline"]
-        runs_b = collated[0]["t::b"][0]["baseline"]
-        assert runs_a == [3, 1, 1, 0]
-        assert runs_b == [3, 2, 0, 2]
+        a = data.tests["t::a"].runs["baseline"]
+        b = data.tests["t::b"].runs["baseline"]
+        assert (a.total, a.failures, a.first_fail, a.first_pass) == \
+            (3, 1, 1, 0)
+        assert (b.total, b.failures, b.first_fail, b.first_pass) == \
+            (3, 2, 0, 2)
 
     def test_outcome_substring_failed(self):
         # the contract is substring matching: "xfailed" counts as failed
-        collated = _runs([["xfailed\tt::a"]], "shuffle")
-        assert collated[0]["t::a"][0]["shuffle"][1] == 1
+        data = _ingest_runs([["xfailed\tt::a"]], "shuffle")
+        assert data.tests["t::a"].runs["shuffle"].failures == 1
 
     def test_nodeid_with_tabs_preserved(self):
         # only the FIRST tab splits outcome from nodeid
-        fd = io.StringIO("passed\tt::a[x\ty]")
-        collated_proj = [{}, None, None, None]
-        update_collated_runs(fd, "baseline", 0, collated_proj)
-        assert "t::a[x\ty]" in collated_proj[0]
+        data = ProjectData("p")
+        data.add_outcomes(io.StringIO("passed\tt::a[x\ty]"), "baseline", 0)
+        assert "t::a[x\ty]" in data.tests
 
 
 class TestLabeling:
     @pytest.mark.parametrize("baseline,shuffle,expect", [
         # incomplete counts -> dropped
-        ([3, 0, None, 0], [4, 0, None, 0], (0, None)),
-        ([4, 0, None, 0], [3, 0, None, 0], (0, None)),
+        ((3, 0, None, 0), (4, 0, None, 0), (0, None)),
+        ((4, 0, None, 0), (3, 0, None, 0), (0, None)),
         # never fails anywhere -> non-flaky
-        ([4, 0, None, 0], [4, 0, None, 0], (0, NON_FLAKY)),
+        ((4, 0, None, 0), (4, 0, None, 0), (0, NON_FLAKY)),
         # baseline clean, shuffle fails -> OD, req = first failing shuffle run
-        ([4, 0, None, 0], [4, 2, 1, 0], (1, OD_FLAKY)),
+        ((4, 0, None, 0), (4, 2, 1, 0), (1, OD_FLAKY)),
         # always fails everywhere -> non-flaky
-        ([4, 4, 0, None], [4, 4, 0, None], (0, NON_FLAKY)),
+        ((4, 4, 0, None), (4, 4, 0, None), (0, NON_FLAKY)),
         # always fails in baseline, shuffle passes once -> OD, req = first pass
-        ([4, 4, 0, None], [4, 3, 0, 2], (2, OD_FLAKY)),
+        ((4, 4, 0, None), (4, 3, 0, 2), (2, OD_FLAKY)),
         # intermittent baseline -> NOD, req = max(first fail, first pass)
-        ([4, 2, 1, 0], [4, 0, None, 0], (1, FLAKY)),
-        ([4, 1, 3, 0], [4, 4, 0, None], (3, FLAKY)),
+        ((4, 2, 1, 0), (4, 0, None, 0), (1, FLAKY)),
+        ((4, 1, 3, 0), (4, 4, 0, None), (3, FLAKY)),
     ])
     def test_decision_table(self, baseline, shuffle, expect):
-        runs = {"baseline": baseline, "shuffle": shuffle}
-        assert get_req_runs_label(runs, N_RUNS_SMALL) == expect
+        runs = {"baseline": _stats(*baseline), "shuffle": _stats(*shuffle)}
+        assert classify(runs, N_RUNS_SMALL) == expect
+
+    def test_missing_mode_is_incomplete(self):
+        assert classify({"baseline": _stats(4, 0, None, 0)},
+                        N_RUNS_SMALL) == (0, None)
 
 
 class TestCoverageFeatures:
@@ -90,10 +101,9 @@ class TestCoverageFeatures:
 
 class TestRusage:
     def test_parse(self):
-        fd = io.StringIO("1.5\t2\t3\t4\t5\t6.25\tt::a")
-        collated_proj = [{}, None, None, None]
-        update_collated_rusage(fd, collated_proj)
-        assert collated_proj[0]["t::a"][2] == [1.5, 2.0, 3.0, 4.0, 5.0, 6.25]
+        data = ProjectData("p")
+        data.add_rusage(io.StringIO("1.5\t2\t3\t4\t5\t6.25\tt::a"))
+        assert data.tests["t::a"].rusage == [1.5, 2.0, 3.0, 4.0, 5.0, 6.25]
 
 
 class TestSyntheticAndLoad:

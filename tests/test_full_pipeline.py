@@ -68,7 +68,7 @@ N_RUNS_SMALL = {"baseline": 6, "shuffle": 12, "testinspect": 1}
 @pytest.mark.timeout(600)
 def test_run_tests_pipeline(tmp_path, monkeypatch):
     from flake16_framework_amd.constants import FLAKY, NON_FLAKY, OD_FLAKY
-    from flake16_framework_amd.dataset.collate import get_collated
+    from flake16_framework_amd.dataset.collate import collate
     from flake16_framework_amd.dataset.tests_io import build_tests
     import flake16_framework_amd.dataset.labeling as labeling
     from flake16_framework_amd.orchestrate import runner
@@ -101,7 +101,7 @@ def test_run_tests_pipeline(tmp_path, monkeypatch):
 
     # --- run stage (resumable driver, process pool, local runner) --------
     with pytest.raises(SystemExit) as exc:
-        runner.run_experiment(
+        runner.drive_runs(
             "baseline", "shuffle", "testinspect",
             subjects_file=str(subjects_file), n_runs=N_RUNS_SMALL)
     assert exc.value.code == 0, open(
@@ -115,10 +115,10 @@ def test_run_tests_pipeline(tmp_path, monkeypatch):
     assert len([f for f in data_files if f.endswith(".tsv")]) >= 6 + 12
 
     # --- collation + labeling -> tests.json ------------------------------
-    collated = get_collated(data_dir=str(tmp_path / "data"),
-                            subjects_dir=str(subjects_dir))
+    collated = collate(data_dir=str(tmp_path / "data"),
+                       subjects_dir=str(subjects_dir))
     tests = build_tests(collated)
-    assert "proj" in tests, list(collated["proj"][0])
+    assert "proj" in tests, list(collated["proj"].tests)
     rows = tests["proj"]
 
     by_name = {nid.split("::")[-1]: row for nid, row in rows.items()}

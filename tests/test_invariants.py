@@ -4,7 +4,8 @@ must hold for ANY input, complementing the example-based suites."""
 import numpy as np
 
 from flake16_framework_amd.constants import FLAKY, NON_FLAKY, OD_FLAKY
-from flake16_framework_amd.dataset.labeling import get_req_runs_label
+from flake16_framework_amd.dataset.collate import RunStats
+from flake16_framework_amd.dataset.labeling import classify
 from flake16_framework_amd.models.binning import bin_codes, compute_bin_cuts
 from flake16_framework_amd.models.forest_ref import (
     LEAF, ForestParams, fit_forest,
@@ -18,24 +19,28 @@ class TestLabelingInvariants:
         for _ in range(500):
             stats = {}
             for mode in ("baseline", "shuffle"):
-                n = int(rng.choice([20, rng.randint(0, 21)]))
-                fails = int(rng.randint(0, n + 1)) if n else 0
-                first_fail = int(rng.randint(0, 20)) if fails else None
-                first_pass = int(rng.randint(0, 20)) if fails < n else None
-                stats[mode] = [n, fails, first_fail, first_pass]
-            req, label = get_req_runs_label(stats, n_runs)
+                st = RunStats()
+                st.total = int(rng.choice([20, rng.randint(0, 21)]))
+                st.failures = int(rng.randint(0, st.total + 1)) \
+                    if st.total else 0
+                st.first_fail = int(rng.randint(0, 20)) \
+                    if st.failures else None
+                st.first_pass = int(rng.randint(0, 20)) \
+                    if st.failures < st.total else None
+                stats[mode] = st
+            req, label = classify(stats, n_runs)
 
             b, s = stats["baseline"], stats["shuffle"]
-            if b[0] != 20 or s[0] != 20:
+            if b.total != 20 or s.total != 20:
                 assert label is None and req == 0
                 continue
             assert label in (NON_FLAKY, OD_FLAKY, FLAKY)
-            if 0 < b[1] < b[0]:
+            if 0 < b.failures < b.total:
                 assert label == FLAKY            # intermittent baseline
-                assert req == max(b[2], b[3])
+                assert req == max(b.first_fail, b.first_pass)
             elif label == NON_FLAKY:
-                assert (b[1] == 0 and s[1] == 0) or \
-                       (b[1] == b[0] and s[1] == s[0])
+                assert (b.failures == 0 and s.failures == 0) or \
+                       (b.failures == b.total and s.failures == s.total)
                 assert req == 0
             else:
                 assert label == OD_FLAKY

@@ -21,39 +21,42 @@ class TestOrchestrate:
                      "bob/proj-b,def456,src,cp x y,pytest -q\n")
         return str(p)
 
-    def test_iter_subjects(self, tmp_path):
-        from flake16_framework_amd.orchestrate.runner import iter_subjects
-        rows = list(iter_subjects(self._subjects_file(tmp_path)))
-        assert rows[0] == ("proj-a", "alice/proj-a", "abc123", ".", "pytest")
-        assert rows[1][0] == "proj-b"
-        assert rows[1][4:] == ("cp x y", "pytest -q")
+    def test_read_subjects(self, tmp_path):
+        from flake16_framework_amd.orchestrate.runner import read_subjects
+        rows = read_subjects(self._subjects_file(tmp_path))
+        assert rows[0] == ("proj-a", "alice/proj-a", "abc123", ".",
+                           ("pytest",))
+        assert rows[1].proj == "proj-b"
+        assert rows[1].commands == ("cp x y", "pytest -q")
 
-    def test_iter_containers_counts(self, tmp_path):
-        from flake16_framework_amd.orchestrate.runner import iter_containers
+    def test_enumerate_runs_counts(self, tmp_path):
+        from flake16_framework_amd.orchestrate.runner import enumerate_runs
         n_runs = {"baseline": 3, "shuffle": 2, "testinspect": 1}
-        conts = list(iter_containers(["baseline", "shuffle"],
-                                     self._subjects_file(tmp_path), n_runs))
+        conts = list(enumerate_runs(["baseline", "shuffle"],
+                                    self._subjects_file(tmp_path), n_runs))
         assert len(conts) == 2 * (3 + 2)
         names = {c for c, _ in conts}
         assert "proj-a_baseline_0" in names
         assert "proj-b_shuffle_1" in names
 
-    def test_mode_flags(self):
-        from flake16_framework_amd.orchestrate.runner import mode_flags
-        assert mode_flags("baseline", "/d/x") == ["--record-file=/d/x.tsv"]
-        assert "--shuffle" in mode_flags("shuffle", "/d/x")
-        assert mode_flags("testinspect", "/d/x") == ["--testinspect=/d/x"]
+    def test_collector_flags(self):
+        from flake16_framework_amd.orchestrate.runner import collector_flags
+        assert collector_flags("baseline", "/d/x") == \
+            ["--record-file=/d/x.tsv"]
+        assert "--shuffle" in collector_flags("shuffle", "/d/x")
+        assert collector_flags("testinspect", "/d/x") == \
+            ["--testinspect=/d/x"]
 
-    def test_manage_container_invocation(self, tmp_path):
-        from flake16_framework_amd.orchestrate.runner import manage_container
+    def test_exec_suite_invocation(self, tmp_path):
+        from flake16_framework_amd.orchestrate.runner import exec_suite
         calls = []
 
         def fake_run(argv, **kwargs):
             calls.append((argv, kwargs))
 
-        manage_container("proj-a_shuffle_7", "echo pre", "pytest -q",
-                         subjects_dir=str(tmp_path), data_dir=str(tmp_path),
-                         run=fake_run)
+        exec_suite("proj-a_shuffle_7", "echo pre", "pytest -q",
+                   subjects_dir=str(tmp_path), data_dir=str(tmp_path),
+                   run=fake_run)
         assert calls[0][0] == ["echo", "pre"]
         final = calls[1][0]
         assert final[:2] == ["pytest", "-q"]
@@ -144,16 +147,13 @@ class TestCollectPlugins:
         assert proc.returncode == 0, proc.stdout + proc.stderr
 
         # sqlite3: the collation layer can ingest it
-        from flake16_framework_amd.dataset.collate import (
-            get_test_data_nid, update_collated_cov,
-        )
-        collated_proj = [{}, None, None, None]
+        from flake16_framework_amd.dataset.collate import ProjectData
+        data = ProjectData("x", subjects_dir=str(tmp_path))
         with sqlite3.connect(f"{prefix}.sqlite3") as con:
-            update_collated_cov(con, "x", collated_proj,
-                                subjects_dir=str(tmp_path))
-        assert any("test_pass" in nid for nid in collated_proj[0])
-        cov = next(v[1] for k, v in collated_proj[0].items()
-                   if "test_pass" in k)
+            data.add_coverage_db(con)
+        assert any("test_pass" in nid for nid in data.tests)
+        cov = next(rec.coverage for nid, rec in data.tests.items()
+                   if "test_pass" in nid)
         assert any(lines for lines in cov.values())
 
         # rusage tsv: 6 floats per test
@@ -341,12 +341,12 @@ def _pool_task(x):
     return f"done: {x}", x * 2
 
 
-class TestManagePool:
+class TestPooledProgress:
     def test_progress_and_results(self, capsys):
         from multiprocessing import Pool
-        from flake16_framework_amd.orchestrate.runner import manage_pool
+        from flake16_framework_amd.orchestrate.runner import pooled_progress
         with Pool(2) as pool:
-            results = sorted(manage_pool(pool, _pool_task, [1, 2, 3]))
+            results = sorted(pooled_progress(pool, _pool_task, [1, 2, 3]))
         assert results == [2, 4, 6]
         out = capsys.readouterr().out
         assert "done:" in out and "3/0" in out
