@@ -79,13 +79,25 @@ class _LineTracer:
 
 
 # ---------------------------------------------------------------------------
-# static metrics (ast-based)
+# static metrics (ast-based, radon-rule implementations)
+#
+# The original study computed Halstead Volume / Cyclomatic Complexity /
+# Maintainability via radon 5.1.0 (reference requirements.txt:25).  radon
+# is not installable here; these follow radon's documented rules:
+#   CC: +1 per if/elif/ternary/assert/with; +1 per loop plus +1 for a
+#       loop else; per try: +1 per except handler plus +1 for try-else;
+#       +1 per comprehension generator plus +1 per condition in it;
+#       boolean op chains add (#values - 1).
+#   Halstead: operators are the arithmetic/boolean/comparison operator
+#       instances (BinOp/UnaryOp/BoolOp/AugAssign/Compare ops); operands
+#       are their direct Name/Constant/Attribute operands.
+#       Volume = (N1 + N2) * log2(n1 + n2).
+#   MI: max(0, (171 - 5.2 ln V - 0.23 CC - 16.2 ln LOC) * 100 / 171)
+#       (the comment term is 0: radon's sin-of-comment-ratio applies to
+#       commented code only).
+# Fixture tests with hand-computed values: tests/test_static_metrics.py.
 # ---------------------------------------------------------------------------
-_BRANCH_NODES = (ast.If, ast.For, ast.While, ast.IfExp, ast.ExceptHandler,
-                 ast.With, ast.Assert, ast.BoolOp, ast.comprehension)
-_OPERATOR_NODES = (ast.BinOp, ast.UnaryOp, ast.BoolOp, ast.Compare,
-                   ast.Call, ast.Subscript, ast.Attribute, ast.Assign,
-                   ast.AugAssign, ast.Return)
+_HAL_OPERAND_TYPES = (ast.Name, ast.Constant, ast.Attribute)
 
 
 def _ast_depth(node, depth=0):
@@ -95,15 +107,42 @@ def _ast_depth(node, depth=0):
     return max(_ast_depth(c, depth + 1) for c in children)
 
 
+def _hal_operand_name(n):
+    if isinstance(n, ast.Name):
+        return n.id
+    if isinstance(n, ast.Attribute):
+        return n.attr
+    return repr(n.value)
+
+
 def _halstead_volume(fn_node):
     operators, operands = [], []
+
+    def operand(n):
+        if isinstance(n, _HAL_OPERAND_TYPES):
+            operands.append(_hal_operand_name(n))
+
     for n in ast.walk(fn_node):
-        if isinstance(n, _OPERATOR_NODES):
-            operators.append(type(n).__name__)
-        elif isinstance(n, ast.Name):
-            operands.append(n.id)
-        elif isinstance(n, ast.Constant):
-            operands.append(repr(n.value))
+        if isinstance(n, ast.BinOp):
+            operators.append(type(n.op).__name__)
+            operand(n.left)
+            operand(n.right)
+        elif isinstance(n, ast.UnaryOp):
+            operators.append(type(n.op).__name__)
+            operand(n.operand)
+        elif isinstance(n, ast.BoolOp):
+            operators.extend([type(n.op).__name__] * (len(n.values) - 1))
+            for v in n.values:
+                operand(v)
+        elif isinstance(n, ast.AugAssign):
+            operators.append(type(n.op).__name__)
+            operand(n.target)
+            operand(n.value)
+        elif isinstance(n, ast.Compare):
+            operators.extend(type(op).__name__ for op in n.ops)
+            operand(n.left)
+            for c in n.comparators:
+                operand(c)
     n1, n2 = len(set(operators)), len(set(operands))
     N1, N2 = len(operators), len(operands)
     vocab = n1 + n2
@@ -114,10 +153,17 @@ def _halstead_volume(fn_node):
 def _cyclomatic(fn_node):
     cc = 1
     for n in ast.walk(fn_node):
-        if isinstance(n, ast.BoolOp):
-            cc += len(n.values) - 1
-        elif isinstance(n, _BRANCH_NODES):
+        if isinstance(n, (ast.If, ast.IfExp, ast.Assert, ast.With,
+                          ast.AsyncWith)):
             cc += 1
+        elif isinstance(n, (ast.For, ast.AsyncFor, ast.While)):
+            cc += 1 + bool(n.orelse)
+        elif isinstance(n, ast.Try):
+            cc += len(n.handlers) + bool(n.orelse)
+        elif isinstance(n, ast.BoolOp):
+            cc += len(n.values) - 1
+        elif isinstance(n, ast.comprehension):
+            cc += 1 + len(n.ifs)
     return cc
 
 
