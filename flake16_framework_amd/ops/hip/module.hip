@@ -178,47 +178,83 @@ std::vector<at::Tensor> forest_fit(
 
     int GRID = 4096;
     if (const char* e = getenv("FLAKE16_FIT_GRID")) GRID = atoi(e);
-    const int CHUNK = 8;
-    int cur = 0;
-    long lev = 0;
-    bool done = false;
-    while (!done) {
-        for (int c = 0; c < CHUNK; ++c, ++lev) {
-            int nx = cur ^ 1;
-            // one 16 B memset clears next-level work/pool/small/mid counts
-            CHECK_HIP(hipMemsetAsync(st + nx * 4, 0, 16, stream));
-            a.sidx_cur = (cur == 0 ? sidx_a : sidx_b).data_ptr<int>();
-            a.sidx_nxt = (cur == 0 ? sidx_b : sidx_a).data_ptr<int>();
-            a.cur = (const WorkItem*)(cur == 0 ? work_a : work_b).data_ptr();
-            a.nxt = (WorkItem*)(cur == 0 ? work_b : work_a).data_ptr();
-            a.cur_count = st + cur * 4;
-            a.nxt_count = st + nx * 4;
-            a.small_count = st + nx * 4 + 2;
-            a.mid_count = st + nx * 4 + 3;
-            if (splitter_random) {
-                et_split_kernel<<<GRID, HBLK, 0, stream>>>(a);
-            } else if (!has_wide && max_features < F &&
-                       getenv("FLAKE16_RF_CANDONLY")) {
-                // ablation variant: candidate-only RF histograms measured
-                // ~3% SLOWER than full histograms + subtraction pools
-                rf_cand_split_kernel<<<GRID, HBLK, 0, stream>>>(a);
-            } else {
-                hist_split_kernel<false><<<GRID, HBLK, 0, stream>>>(a);
-                if (has_wide)
-                    hist_split_kernel<true><<<GRID, HBLK, 0, stream>>>(a);
-            }
-            mid_subtree_kernel<<<2048, HBLK, 0, stream>>>(a, a.sidx_nxt);
-            small_subtree_kernel<<<2048, HBLK, 0, stream>>>(
-                a, a.sidx_nxt);
-            CHECK_HIP(hipMemcpyAsync(pinned_p + (c % PINSZ),
-                                     st + nx * 4, 4,
-                                     hipMemcpyDeviceToHost, stream));
-            cur = nx;
+
+    // One level's dispatches: clear next-parity counters, split kernel(s),
+    // mid- and small-subtree drains, next-level count -> pinned slot.
+    auto level_ops = [&](int cur_par, int pinned_slot) {
+        const int nx = cur_par ^ 1;
+        // one 16 B memset clears next-level work/pool/small/mid counts
+        CHECK_HIP(hipMemsetAsync(st + nx * 4, 0, 16, stream));
+        a.sidx_cur = (cur_par == 0 ? sidx_a : sidx_b).data_ptr<int>();
+        a.sidx_nxt = (cur_par == 0 ? sidx_b : sidx_a).data_ptr<int>();
+        a.cur = (const WorkItem*)(cur_par == 0 ? work_a : work_b).data_ptr();
+        a.nxt = (WorkItem*)(cur_par == 0 ? work_b : work_a).data_ptr();
+        a.cur_count = st + cur_par * 4;
+        a.nxt_count = st + nx * 4;
+        a.small_count = st + nx * 4 + 2;
+        a.mid_count = st + nx * 4 + 3;
+        if (splitter_random) {
+            et_split_kernel<<<GRID, HBLK, 0, stream>>>(a);
+        } else if (!has_wide && max_features < F &&
+                   getenv("FLAKE16_RF_CANDONLY")) {
+            // ablation variant: candidate-only RF histograms measured
+            // ~3% SLOWER than full histograms + subtraction pools
+            rf_cand_split_kernel<<<GRID, HBLK, 0, stream>>>(a);
+        } else {
+            hist_split_kernel<false><<<GRID, HBLK, 0, stream>>>(a);
+            if (has_wide)
+                hist_split_kernel<true><<<GRID, HBLK, 0, stream>>>(a);
         }
-        CHECK_HIP(hipStreamSynchronize(stream));
-        for (int c = 0; c < CHUNK; ++c)
-            if (pinned_p[c] == 0) { done = true; break; }
-        TORCH_CHECK(lev < 8192, "forest_fit: depth limit exceeded");
+        mid_subtree_kernel<<<2048, HBLK, 0, stream>>>(a, a.sidx_nxt);
+        small_subtree_kernel<<<2048, HBLK, 0, stream>>>(a, a.sidx_nxt);
+        CHECK_HIP(hipMemcpyAsync(pinned_p + pinned_slot, st + nx * 4, 4,
+                                 hipMemcpyDeviceToHost, stream));
+    };
+
+    if (!getenv("FLAKE16_NO_HIPGRAPH")) {
+        // The dispatch sequence is parity-periodic with every pointer
+        // fixed, so ONE even+odd level pair is captured into a hipGraph
+        // and replayed: ~12 host dispatches per level pair collapse into
+        // one graph launch (the host-side submission lock was measured
+        // limiting multi-stream overlap).  Replaying past exhaustion is
+        // harmless — all kernels early-exit on zero counts.
+        hipGraph_t graph = nullptr;
+        hipGraphExec_t gexec = nullptr;
+        CHECK_HIP(hipStreamBeginCapture(stream,
+                                        hipStreamCaptureModeThreadLocal));
+        level_ops(0, 0);
+        level_ops(1, 1);
+        CHECK_HIP(hipStreamEndCapture(stream, &graph));
+        CHECK_HIP(hipGraphInstantiate(&gexec, graph, nullptr, nullptr, 0));
+
+        const int PAIRS_PER_SYNC = 4;
+        long pairs = 0;
+        bool done = false;
+        while (!done) {
+            for (int p = 0; p < PAIRS_PER_SYNC; ++p)
+                CHECK_HIP(hipGraphLaunch(gexec, stream));
+            CHECK_HIP(hipStreamSynchronize(stream));
+            pairs += PAIRS_PER_SYNC;
+            done = pinned_p[0] == 0 || pinned_p[1] == 0;
+            TORCH_CHECK(pairs < 4096, "forest_fit: depth limit exceeded");
+        }
+        CHECK_HIP(hipGraphExecDestroy(gexec));
+        CHECK_HIP(hipGraphDestroy(graph));
+    } else {
+        const int CHUNK = 8;
+        int cur = 0;
+        long lev = 0;
+        bool done = false;
+        while (!done) {
+            for (int c = 0; c < CHUNK; ++c, ++lev) {
+                level_ops(cur, c % PINSZ);
+                cur ^= 1;
+            }
+            CHECK_HIP(hipStreamSynchronize(stream));
+            for (int c = 0; c < CHUNK; ++c)
+                if (pinned_p[c] == 0) { done = true; break; }
+            TORCH_CHECK(lev < 8192, "forest_fit: depth limit exceeded");
+        }
     }
 
     TORCH_CHECK(err.to(at::kCPU).item<int>() == 0,
