@@ -211,13 +211,16 @@ std::vector<at::Tensor> forest_fit(
                                  hipMemcpyDeviceToHost, stream));
     };
 
-    if (!getenv("FLAKE16_NO_HIPGRAPH")) {
-        // The dispatch sequence is parity-periodic with every pointer
-        // fixed, so ONE even+odd level pair is captured into a hipGraph
-        // and replayed: ~12 host dispatches per level pair collapse into
-        // one graph launch (the host-side submission lock was measured
-        // limiting multi-stream overlap).  Replaying past exhaustion is
-        // harmless — all kernels early-exit on zero counts.
+    if (getenv("FLAKE16_HIPGRAPH") && stream != nullptr) {
+        // MEASURED AND REJECTED (kept for A/B): the dispatch sequence is
+        // parity-periodic with every pointer fixed, so one even+odd level
+        // pair is captured into a hipGraph and replayed — but the
+        // per-fit capture+instantiate costs more than the ~12 dispatches
+        // per pair it saves (same-box sweep 123.9 vs 127.1 configs/s),
+        // so the plain loop below is the default.  Requires a non-default
+        // stream (capture on the legacy stream is not permitted).
+        // Replaying past exhaustion is harmless — all kernels early-exit
+        // on zero counts.
         hipGraph_t graph = nullptr;
         hipGraphExec_t gexec = nullptr;
         CHECK_HIP(hipStreamBeginCapture(stream,
