@@ -389,22 +389,26 @@ class SweepContext:
 
         fold_codes = [c for c, _ in balanced]
         fold_labels = [y for _, y in balanced]
-        j_row_off, j_n, j_key = [], [], []
-        row_base = 0
-        for i in range(N_FOLDS):
-            _, job_base = job_ids_for(config_keys, cell_idx, i)
-            n_i = fold_labels[i].shape[0]
-            for t in range(n_trees):
-                j_row_off.append(row_base)
-                j_n.append(n_i)
-                j_key.append(job_base + t)
-            row_base += n_i
+        # vectorized job tables (a python loop here is 10*n_trees items per
+        # cell of pure-GIL time, which serializes the 4 worker threads)
+        n_per_fold = np.array([int(y.shape[0]) for y in fold_labels],
+                              dtype=np.int64)
+        fold_base = np.concatenate(([0], np.cumsum(n_per_fold)[:-1]))
+        job_bases = np.array(
+            [job_ids_for(config_keys, cell_idx, i)[1]
+             for i in range(N_FOLDS)], dtype=np.int64)
+        j_row_off_np = np.repeat(fold_base, n_trees).astype(np.int32)
+        j_n_np = np.repeat(n_per_fold, n_trees).astype(np.int32)
+        j_key_np = (np.repeat(job_bases, n_trees)
+                    + np.tile(np.arange(n_trees), N_FOLDS)).astype(np.int32)
 
         codes_train = torch.cat(fold_codes, dim=0).contiguous()
         labels_train = torch.cat(fold_labels, dim=0).contiguous()
-        j_row_off = torch.tensor(j_row_off, dtype=torch.int32, device=device)
-        j_n = torch.tensor(j_n, dtype=torch.int32, device=device)
-        j_key = torch.tensor(j_key, dtype=torch.int32, device=device)
+        j_row_off = torch.from_numpy(j_row_off_np).to(device)
+        # j_n stays CPU-resident: forest_fit needs the values host-side
+        # for workspace sizing (a device tensor would force a D2H sync)
+        j_n = torch.from_numpy(j_n_np)
+        j_key = torch.from_numpy(j_key_np).to(device)
 
         ev = [torch.cuda.Event(enable_timing=True) for _ in range(4)]
         ev[0].record()

@@ -1459,62 +1459,16 @@ __global__ void mid_subtree_kernel(ForestDev a,
             m_idx[i] = (uint16_t)i;
         }
 
-        if (a.wave_mid) {
-            // Wave-parallel rounds: up to HBLK/64 nodes in flight, each
-            // built by one wave (no block barriers inside a node).  The
-            // block-wide 16 KiB `hist` array is re-sliced as one
-            // WAVE_CANDS*256-word histogram region per wave.
-            if (tid == 0) {
-                sh_count = 1;
-                stack[0] = {0, (short)n0, it.depth, it.node};
-            }
-            while (true) {
-                __syncthreads();
-                if (tid == 0) {
-                    const int take = min(HBLK / 64, sh_count);
-                    for (int w = 0; w < take; ++w)
-                        wframes[w] = stack[sh_count - 1 - w];
-                    sh_count -= take;
-                    sh_take = take;
-                }
-                __syncthreads();
-                if (sh_take == 0) break;
-                if (wave < sh_take) {
-                    const MidFrame fr = wframes[wave];
-                    mid_wave_node(a, it, fr.s, fr.e, fr.depth, fr.node,
-                                  m_codes, m_lab, m_idx, m_idx2,
-                                  hist + wave * (WAVE_CANDS * 256),
-                                  wperm_ws[wave], wthr_ws[wave],
-                                  wcand_ws[wave], wcbin_ws[wave],
-                                  stack, &sh_count);
-                }
-            }
-            __syncthreads();
-            for (int i = tid; i < n0; i += HBLK)
-                a.sidx_nxt[sbase + it.start + i] = m_rows[m_idx[i]];
-            __syncthreads();
-            continue;
-        }
-
-        if (tid == 0) {
-            sh_sp = 0;
-            stack[0] = {0, (short)n0, it.depth, it.node};
-        }
-        __syncthreads();
-
-        while (true) {
-            const int sp = sh_sp;
-            if (sp < 0) break;
-            const int ls = stack[sp].s;
-            const int le = stack[sp].e;
-            const int depth = stack[sp].depth;
-            const int node = stack[sp].node;
+        // Per-node block-wide builder (histogram, split, partition over
+        // the staged window) shared by the DFS path and the wave mode's
+        // single-frontier rounds.  dfs_mode selects the child routing:
+        // the DFS stack (smaller child on top) or the wave stack.
+        auto block_node = [&](int ls, int le, int depth, int node,
+                              bool dfs_mode) {
             const int n = le - ls;
             // global job-relative range (the RNG identity)
             const int gs = it.start + ls;
             const int ge = it.start + le;
-            __syncthreads();
-            if (tid == 0) --sh_sp;
 
             // histogram from LDS-resident samples
             for (int i = tid; i < F * 64; i += HBLK)
@@ -1549,7 +1503,7 @@ __global__ void mid_subtree_kernel(ForestDev a,
             }
             if (n < 2 || c0 == 0 || c1 == 0) {
                 __syncthreads();
-                continue;
+                return;
             }
 
             // occupied-bin range per feature
@@ -1603,7 +1557,7 @@ __global__ void mid_subtree_kernel(ForestDev a,
             const int ncand = sh_ncand;
             if (ncand == 0) {
                 __syncthreads();
-                continue;
+                return;
             }
 
             // evaluate candidates (wave per candidate; identical math and
@@ -1718,7 +1672,7 @@ __global__ void mid_subtree_kernel(ForestDev a,
             const int bf = sh_bestf;
             if (bf < 0) {
                 __syncthreads();
-                continue;
+                return;
             }
             const int bb = sh_bestbin;
             const int nL = sh_bestnL;
@@ -1765,9 +1719,6 @@ __global__ void mid_subtree_kernel(ForestDev a,
             for (int i = ls + tid; i < le; i += HBLK) m_idx[i] = m_idx2[i];
             __syncthreads();
 
-            // route children: > SMALL_N continue in-block (smaller child
-            // processed first: push larger first); <= SMALL_N go to the
-            // small queue with their final global range
             if (tid == 0) {
                 const int nR = n - nL;
                 const MidFrame left = {(short)ls, (short)(ls + nL),
@@ -1782,9 +1733,7 @@ __global__ void mid_subtree_kernel(ForestDev a,
                 for (int x = 0; x < 2; ++x) {
                     const MidFrame& fr = *ordered[x];
                     const int fn = fr.e - fr.s;
-                    if (fn > SMALL_N) {
-                        stack[++sh_sp] = fr;
-                    } else {
+                    if (fn <= SMALL_N) {
                         const int i = atomicAdd(a.small_count, 1);
                         if (i < a.small_cap)
                             a.small[i] = {it.job, fr.node,
@@ -1792,10 +1741,78 @@ __global__ void mid_subtree_kernel(ForestDev a,
                                           fr.depth, -1};
                         else
                             atomicExch(a.err_flag, 1);
+                    } else if (dfs_mode) {
+                        stack[++sh_sp] = fr;
+                    } else if (sh_count < MID_STACK) {
+                        stack[sh_count++] = fr;
+                    } else {
+                        const int i2 = atomicAdd(a.nxt_count, 1);
+                        if (i2 < a.work_cap)
+                            a.nxt[i2] = {it.job, fr.node, it.start + fr.s,
+                                         it.start + fr.e, fr.depth, -1};
+                        else
+                            atomicExch(a.err_flag, 1);
                     }
                 }
             }
             __syncthreads();
+        };
+
+        if (a.wave_mid) {
+            // Wave-parallel rounds: up to HBLK/64 nodes in flight, each
+            // built by one wave (no block barriers inside a node); a
+            // single-node frontier — the top of every staged subtree —
+            // is built block-wide instead, so no wave idles there.  The
+            // block-wide 16 KiB `hist` array is re-sliced as one
+            // WAVE_CANDS*256-word histogram region per wave.
+            if (tid == 0) {
+                sh_count = 1;
+                stack[0] = {0, (short)n0, it.depth, it.node};
+            }
+            while (true) {
+                __syncthreads();
+                if (tid == 0) {
+                    const int take = min(HBLK / 64, sh_count);
+                    for (int w = 0; w < take; ++w)
+                        wframes[w] = stack[sh_count - 1 - w];
+                    sh_count -= take;
+                    sh_take = take;
+                }
+                __syncthreads();
+                if (sh_take == 0) break;
+                if (sh_take == 1) {
+                    const MidFrame fr = wframes[0];
+                    block_node(fr.s, fr.e, fr.depth, fr.node, false);
+                } else if (wave < sh_take) {
+                    const MidFrame fr = wframes[wave];
+                    mid_wave_node(a, it, fr.s, fr.e, fr.depth, fr.node,
+                                  m_codes, m_lab, m_idx, m_idx2,
+                                  hist + wave * (WAVE_CANDS * 256),
+                                  wperm_ws[wave], wthr_ws[wave],
+                                  wcand_ws[wave], wcbin_ws[wave],
+                                  stack, &sh_count);
+                }
+            }
+            __syncthreads();
+            for (int i = tid; i < n0; i += HBLK)
+                a.sidx_nxt[sbase + it.start + i] = m_rows[m_idx[i]];
+            __syncthreads();
+            continue;
+        }
+
+        if (tid == 0) {
+            sh_sp = 0;
+            stack[0] = {0, (short)n0, it.depth, it.node};
+        }
+        __syncthreads();
+
+        while (true) {
+            const int sp = sh_sp;
+            if (sp < 0) break;
+            const MidFrame fr = stack[sp];
+            __syncthreads();
+            if (tid == 0) --sh_sp;
+            block_node(fr.s, fr.e, fr.depth, fr.node, true);
         }
 
         // write the final arrangement back so the small kernel (launched
@@ -1805,6 +1822,7 @@ __global__ void mid_subtree_kernel(ForestDev a,
             a.sidx_nxt[sbase + it.start + i] = m_rows[m_idx[i]];
         __syncthreads();
     }
+}
 }
 
 // ---------------------------------------------------------------------------

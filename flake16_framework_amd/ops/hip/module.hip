@@ -46,7 +46,10 @@ std::vector<at::Tensor> forest_fit(
     auto opts_i64 = codes.options().dtype(at::kLong);
     auto opts_f32 = codes.options().dtype(at::kFloat);
 
-    auto j_n_cpu = j_n.to(at::kCPU);
+    // j_n may arrive CPU-resident (preferred: the host needs its values
+    // for sizing, and a device tensor would force a blocking D2H here)
+    auto j_n_cpu = j_n.is_cuda() ? j_n.to(at::kCPU) : j_n;
+    auto j_n_dev = j_n.is_cuda() ? j_n : j_n.to(codes.device());
     const int* jn = j_n_cpu.data_ptr<int>();
     std::vector<long> sidx_off(J), node_off(J);
     long S = 0, Ntot = 0;
@@ -129,7 +132,7 @@ std::vector<at::Tensor> forest_fit(
     hipStream_t stream = current_stream();
 
     forest_init_kernel<<<J, HBLK, 0, stream>>>(
-        j_row_off.data_ptr<int>(), j_n.data_ptr<int>(),
+        j_row_off.data_ptr<int>(), j_n_dev.data_ptr<int>(),
         j_sidx_off.data_ptr<long>(), j_key.data_ptr<int>(),
         node_alloc.data_ptr<int>(), sidx_a.data_ptr<int>(),
         (WorkItem*)work_a.data_ptr(), bootstrap ? 1 : 0, (uint32_t)seed);
@@ -145,7 +148,7 @@ std::vector<at::Tensor> forest_fit(
     a.codes = codes.data_ptr<uint8_t>();
     a.labels = labels.data_ptr<uint8_t>();
     a.j_row_off = j_row_off.data_ptr<int>();
-    a.j_n = j_n.data_ptr<int>();
+    a.j_n = j_n_dev.data_ptr<int>();
     a.j_sidx_off = j_sidx_off.data_ptr<long>();
     a.j_node_off = j_node_off.data_ptr<long>();
     a.j_key = j_key.data_ptr<int>();
