@@ -370,72 +370,115 @@ class SweepContext:
         self._balanced[bk] = out
         return out
 
-    # -- the per-cell evaluation ------------------------------------------
+    # -- evaluation -------------------------------------------------------
     def evaluate_cell(self, config_keys, cell_idx):
+        return self.evaluate_group([(config_keys, cell_idx)])[config_keys]
+
+    def evaluate_group(self, group_cells):
+        """Evaluate 1-3 cells of ONE balance group (same keys[:4]) with a
+        single fused forest_fit_multi over all their jobs — the DT + RF +
+        ET trees of a group share one level pipeline, which cuts kernel
+        launches ~3x and keeps the work queues full.
+
+        Trees are bit-identical to per-cell fits: every job keeps the
+        same Philox key (job_ids_for), and all kernels read the model
+        spec per job.  The fused t_train is attributed to cells in
+        proportion to their tree counts (documented deviation: the
+        reference records per-process wall time, experiment.py:455).
+        Returns {config_keys: [t_train, t_test, scores, scores_total]}.
+        """
         from .scores import job_ids_for
 
         ops, device = self.ops, self.device
-        view = self.view_for(config_keys)
-        lab = self.labels_for(config_keys)
-        balanced = self.balanced_for(config_keys, cell_idx)
+        keys0, cell0 = group_cells[0]
+        view = self.view_for(keys0)
+        lab = self.labels_for(keys0)
+        balanced = self.balanced_for(keys0, cell0)
         F = view["F"]
-
-        spec = MODEL_AXIS[config_keys[4]]
-        n_trees = spec["n_estimators"]
-        bootstrap = spec["bootstrap"]
-        splitter_random = spec["kind"] == "extra_trees"
-        max_features = F if spec["kind"] == "decision_tree" else max(
-            1, int(np.sqrt(F)))
 
         fold_codes = [c for c, _ in balanced]
         fold_labels = [y for _, y in balanced]
-        # vectorized job tables (a python loop here is 10*n_trees items per
-        # cell of pure-GIL time, which serializes the 4 worker threads)
         n_per_fold = np.array([int(y.shape[0]) for y in fold_labels],
                               dtype=np.int64)
         fold_base = np.concatenate(([0], np.cumsum(n_per_fold)[:-1]))
-        job_bases = np.array(
-            [job_ids_for(config_keys, cell_idx, i)[1]
-             for i in range(N_FOLDS)], dtype=np.int64)
-        j_row_off_np = np.repeat(fold_base, n_trees).astype(np.int32)
-        j_n_np = np.repeat(n_per_fold, n_trees).astype(np.int32)
-        j_key_np = (np.repeat(job_bases, n_trees)
-                    + np.tile(np.arange(n_trees), N_FOLDS)).astype(np.int32)
+
+        # vectorized per-cell job blocks, concatenated cell-major
+        parts = {"row": [], "n": [], "key": [], "mf": [], "rand": [],
+                 "boot": []}
+        cell_meta = []   # (keys, n_trees, job_off)
+        job_off = 0
+        for keys, cell_idx in group_cells:
+            spec = MODEL_AXIS[keys[4]]
+            n_trees = spec["n_estimators"]
+            mf = F if spec["kind"] == "decision_tree" else max(
+                1, int(np.sqrt(F)))
+            job_bases = np.array(
+                [job_ids_for(keys, cell_idx, i)[1]
+                 for i in range(N_FOLDS)], dtype=np.int64)
+            n_jobs = N_FOLDS * n_trees
+            parts["row"].append(np.repeat(fold_base, n_trees))
+            parts["n"].append(np.repeat(n_per_fold, n_trees))
+            parts["key"].append(np.repeat(job_bases, n_trees)
+                                + np.tile(np.arange(n_trees), N_FOLDS))
+            parts["mf"].append(np.full(n_jobs, mf))
+            parts["rand"].append(np.full(
+                n_jobs, spec["kind"] == "extra_trees"))
+            parts["boot"].append(np.full(n_jobs, spec["bootstrap"]))
+            cell_meta.append((keys, n_trees, job_off))
+            job_off += n_jobs
 
         codes_train = torch.cat(fold_codes, dim=0).contiguous()
         labels_train = torch.cat(fold_labels, dim=0).contiguous()
-        j_row_off = torch.from_numpy(j_row_off_np).to(device)
-        # j_n stays CPU-resident: forest_fit needs the values host-side
-        # for workspace sizing (a device tensor would force a D2H sync)
-        j_n = torch.from_numpy(j_n_np)
-        j_key = torch.from_numpy(j_key_np).to(device)
+        cat32 = lambda k: np.concatenate(parts[k]).astype(np.int32)
+        j_row_off = torch.from_numpy(cat32("row")).to(device)
+        # j_n / j_mf / j_rand / j_boot stay CPU-resident: forest_fit_multi
+        # needs their values host-side (sizing + kernel selection)
+        j_n = torch.from_numpy(cat32("n"))
+        j_key = torch.from_numpy(cat32("key")).to(device)
+        j_mf = torch.from_numpy(cat32("mf"))
+        j_rand = torch.from_numpy(
+            np.concatenate(parts["rand"]).astype(np.uint8))
+        j_boot = torch.from_numpy(
+            np.concatenate(parts["boot"]).astype(np.uint8))
 
-        ev = [torch.cuda.Event(enable_timing=True) for _ in range(4)]
-        ev[0].record()
+        ev_fit = [torch.cuda.Event(enable_timing=True) for _ in range(2)]
+        ev_fit[0].record()
         nfeat, nsplit, nleft, ncnt0, ncnt1, j_node_off, node_alloc = \
-            ops.forest_fit(codes_train, labels_train, j_row_off, j_n, j_key,
-                           F, max_features, bootstrap, splitter_random,
-                           self.seed)
-        ev[1].record()
+            ops.forest_fit_multi(codes_train, labels_train, j_row_off, j_n,
+                                 j_key, j_mf, j_rand, j_boot, F, self.seed)
+        ev_fit[1].record()
 
-        ev[2].record()
-        pred, confusion = ops.forest_predict_confusion(
-            view["codes_all"], lab["labels_dev"], self._proj_id,
-            lab["pair_row"], lab["pair_fold"], j_node_off,
-            nfeat, nsplit, nleft, ncnt0, ncnt1, n_trees,
-            len(self._proj_uniq))
-        ev[3].record()
-        ev[3].synchronize()
+        out = {}
+        pred_events = []
+        for keys, n_trees, off in cell_meta:
+            e0 = torch.cuda.Event(enable_timing=True)
+            e1 = torch.cuda.Event(enable_timing=True)
+            e0.record()
+            _, confusion = ops.forest_predict_confusion(
+                view["codes_all"], lab["labels_dev"], self._proj_id,
+                lab["pair_row"], lab["pair_fold"],
+                j_node_off.narrow(0, off, N_FOLDS * n_trees),
+                nfeat, nsplit, nleft, ncnt0, ncnt1, n_trees,
+                len(self._proj_uniq))
+            e1.record()
+            pred_events.append((keys, confusion, e0, e1))
 
-        t_train = ev[0].elapsed_time(ev[1]) / 1000.0
-        t_test = ev[2].elapsed_time(ev[3]) / 1000.0
+        pred_events[-1][3].synchronize()
+        t_fit = ev_fit[0].elapsed_time(ev_fit[1]) / 1000.0
+        total_trees = sum(nt for _, nt, _ in cell_meta)
 
-        conf = confusion.cpu().numpy()
-        scores = {p: [int(conf[i, 0]), int(conf[i, 1]), int(conf[i, 2])]
-                  for i, p in enumerate(self._proj_uniq)}
-        scores_total = [int(v) for v in conf[len(self._proj_uniq)]]
-        finalize_scores(scores, scores_total)
-        return [t_train / N_FOLDS, t_test / N_FOLDS, scores, scores_total]
+        for (keys, n_trees, _), (_, confusion, e0, e1) in zip(cell_meta,
+                                                              pred_events):
+            t_train = t_fit * n_trees / total_trees
+            t_test = e0.elapsed_time(e1) / 1000.0
+            conf = confusion.cpu().numpy()
+            scores = {p: [int(conf[i, 0]), int(conf[i, 1]), int(conf[i, 2])]
+                      for i, p in enumerate(self._proj_uniq)}
+            scores_total = [int(v) for v in conf[len(self._proj_uniq)]]
+            finalize_scores(scores, scores_total)
+            out[keys] = [t_train / N_FOLDS, t_test / N_FOLDS, scores,
+                         scores_total]
+        return out
 
 
 def evaluate_cell_hip(config_keys, cell_idx, tests=None, tests_file=None,

@@ -167,15 +167,18 @@ def _eval_cell_ref_task(args):
 def _run_scores_hip(all_cells, tests, tests_file, seed, progress,
                     n_streams=None, on_result=None):
     """Device sweep: prebuild the shared caches (views, folds, balanced
-    groups) on the default stream, then evaluate cells concurrently on
-    worker threads with one HIP stream each — host-side cell bookkeeping
-    overlaps other cells' kernels (extension calls release the GIL)."""
+    groups) on the default stream, then evaluate BALANCE GROUPS (the 1-3
+    model cells sharing balanced folds, fused into one mixed-model
+    forest_fit) concurrently on worker threads with one HIP stream each —
+    host-side bookkeeping overlaps other groups' kernels (extension calls
+    release the GIL)."""
     import os
     import threading
     from concurrent.futures import ThreadPoolExecutor
 
     import torch
 
+    from ..configgrid import balance_group_index
     from ..utils.trace import trace_span
     from .hip_cell import SweepContext
 
@@ -184,35 +187,41 @@ def _run_scores_hip(all_cells, tests, tests_file, seed, progress,
 
     # Shared caches (views, folds, balanced groups) build lazily under
     # per-key locks: the first worker to need an entry builds it on its own
-    # stream while other workers proceed with other cells.
+    # stream while other workers proceed with other groups.
     context = SweepContext(tests=tests, tests_file=tests_file, seed=seed)
+
+    groups = {}
+    for cell_idx, config_keys in all_cells:
+        g = balance_group_index(config_keys)
+        groups.setdefault(g, []).append((config_keys, cell_idx))
+    tasks = [groups[g] for g in sorted(groups)]
 
     out = {}
     lock = threading.Lock()
     t_start = time.time()
     n_done = [0]
 
-    def eval_one(args):
-        cell_idx, config_keys = args
+    def eval_group(group_cells):
         stream = torch.cuda.Stream()
-        with trace_span("cell", cell=cell_idx, backend="hip"):
+        with trace_span("group", cells=len(group_cells), backend="hip"):
             with torch.cuda.stream(stream):
-                result = context.evaluate_cell(config_keys, cell_idx)
+                result = context.evaluate_group(group_cells)
         with lock:
-            out[config_keys] = result
-            n_done[0] += 1
-            if on_result:
-                on_result(config_keys, result)
-            if progress:
-                progress(n_done[0], len(all_cells), time.time() - t_start,
-                         ", ".join(config_keys))
+            for config_keys, value in result.items():
+                out[config_keys] = value
+                n_done[0] += 1
+                if on_result:
+                    on_result(config_keys, value)
+                if progress:
+                    progress(n_done[0], len(all_cells),
+                             time.time() - t_start, ", ".join(config_keys))
 
     if n_streams <= 1:
-        for args in all_cells:
-            eval_one(args)
+        for group_cells in tasks:
+            eval_group(group_cells)
     else:
         with ThreadPoolExecutor(max_workers=n_streams) as pool:
-            list(pool.map(eval_one, all_cells))
+            list(pool.map(eval_group, tasks))
     return out
 
 

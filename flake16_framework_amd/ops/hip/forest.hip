@@ -63,8 +63,10 @@ struct ForestDev {
     int* __restrict__ nxt_count;
     int* __restrict__ err_flag;
     int F;
-    int max_features;
-    int splitter_random;
+    // Per-job model spec (a fused fit batches a balance group's three
+    // model cells: DT + RF + ET jobs in one build).
+    const int* __restrict__ j_mf;        // [J] max_features
+    const uint8_t* __restrict__ j_rand;  // [J] splitter: 1 = random (ET)
     uint32_t seed;
     int work_cap;
     // Histogram-subtraction pools: a splitting node >= hist_save_min
@@ -126,12 +128,14 @@ __global__ void forest_init_kernel(
     const int* __restrict__ j_row_off, const int* __restrict__ j_n,
     const long* __restrict__ j_sidx_off, const int* __restrict__ j_key,
     int* __restrict__ node_alloc, int* __restrict__ sidx,
-    WorkItem* __restrict__ work, int bootstrap, uint32_t seed) {
+    WorkItem* __restrict__ work, const uint8_t* __restrict__ j_boot,
+    uint32_t seed) {
     int job = blockIdx.x;
     int n = j_n[job];
     long off = j_sidx_off[job];
     int row0 = j_row_off[job];
     uint32_t key = (uint32_t)j_key[job];
+    const int bootstrap = j_boot[job];
 
     for (int i = threadIdx.x; i < n; i += blockDim.x) {
         int s;
@@ -186,6 +190,8 @@ __global__ void hist_split_kernel(ForestDev a) {
         WorkItem it = a.cur[wi];
         const int n = it.end - it.start;
         if ((n >= 65536) != WIDE) continue;   // size-class filter (uniform)
+        if (a.j_rand[it.job]) continue;       // random-splitter: et_split's
+        const int max_features = a.j_mf[it.job];
         const long sbase = a.j_sidx_off[it.job];
         const long nbase = a.j_node_off[it.job];
         const uint32_t key = (uint32_t)a.j_key[it.job];
@@ -298,7 +304,7 @@ __global__ void hist_split_kernel(ForestDev a) {
                 int t = perm[i]; perm[i] = perm[j]; perm[j] = t;
             }
             int nc = 0;
-            for (int i = 0; i < F && nc < a.max_features; ++i) {
+            for (int i = 0; i < F && nc < max_features; ++i) {
                 int f = perm[i];
                 if (sh_bmin[f] != sh_bmax[f]) sh_cand[nc++] = f;
             }
@@ -340,7 +346,7 @@ __global__ void hist_split_kernel(ForestDev a) {
             double best_s = -1.0;  // all real scores are > 0
             int best_b = -1, best_nl = 0;
 
-            if (a.splitter_random) {
+            if (a.j_rand[it.job]) {
                 uint32_t tag = TAG_THRESH | ((uint32_t)(it.depth & 0xFF) << 8);
                 uint32_t u = philox_draw(tag, (uint32_t)it.start,
                                          (uint32_t)it.end, (uint32_t)f,
@@ -600,6 +606,8 @@ __global__ void et_split_kernel(ForestDev a) {
     for (int wi = blockIdx.x; wi < n_items; wi += gridDim.x) {
         WorkItem it = a.cur[wi];
         const int n = it.end - it.start;
+        if (!a.j_rand[it.job]) continue;     // best-splitter: hist_split's
+        const int max_features = a.j_mf[it.job];
         const long sbase = a.j_sidx_off[it.job];
         const long nbase = a.j_node_off[it.job];
         const uint32_t key = (uint32_t)a.j_key[it.job];
@@ -677,7 +685,7 @@ __global__ void et_split_kernel(ForestDev a) {
                 int t = perm[i]; perm[i] = perm[j]; perm[j] = t;
             }
             int nc = 0;
-            for (int i = 0; i < F && nc < a.max_features; ++i) {
+            for (int i = 0; i < F && nc < max_features; ++i) {
                 int f = perm[i];
                 if (sh_min[f] == sh_max[f]) continue;
                 sh_cand[nc] = f;
@@ -850,6 +858,8 @@ __global__ void rf_cand_split_kernel(ForestDev a) {
     for (int wi = blockIdx.x; wi < n_items; wi += gridDim.x) {
         WorkItem it = a.cur[wi];
         const int n = it.end - it.start;
+        if (a.j_rand[it.job]) continue;      // random-splitter: et_split's
+        const int max_features = a.j_mf[it.job];
         const long sbase = a.j_sidx_off[it.job];
         const long nbase = a.j_node_off[it.job];
         const uint32_t key = (uint32_t)a.j_key[it.job];
@@ -922,7 +932,7 @@ __global__ void rf_cand_split_kernel(ForestDev a) {
                 int t = perm[i]; perm[i] = perm[j]; perm[j] = t;
             }
             int nc = 0;
-            for (int i = 0; i < F && nc < a.max_features; ++i) {
+            for (int i = 0; i < F && nc < max_features; ++i) {
                 int f = perm[i];
                 if (sh_min[f] != sh_max[f]) sh_cand[nc++] = f;
             }
@@ -1134,6 +1144,8 @@ __device__ __forceinline__ void mid_wave_node(
     const long nbase = a.j_node_off[it.job];
     const uint32_t key = (uint32_t)a.j_key[it.job];
     const int F = a.F;
+    const int max_features = a.j_mf[it.job];
+    const int splitter_random = a.j_rand[it.job];
     const int gs = it.start + ls;   // global job-relative range (RNG id)
     const int ge = it.start + le;
 
@@ -1181,7 +1193,7 @@ __device__ __forceinline__ void mid_wave_node(
             TAG_FEATSEL | ((uint32_t)(depth & 0xFF) << 8),
             (uint32_t)gs, (uint32_t)ge,
             (uint32_t)(lane < FPAD ? lane : 0), a.seed, key);
-        if (a.splitter_random && lane < FPAD)
+        if (splitter_random && lane < FPAD)
             wthr[lane] = philox_draw(
                 TAG_THRESH | ((uint32_t)(depth & 0xFF) << 8),
                 (uint32_t)gs, (uint32_t)ge, (uint32_t)lane, a.seed, key);
@@ -1197,11 +1209,11 @@ __device__ __forceinline__ void mid_wave_node(
     }
     int ncand = 0;
     if (lane == 0) {
-        for (int i = 0; i < F && ncand < a.max_features; ++i) {
+        for (int i = 0; i < F && ncand < max_features; ++i) {
             const int f = wperm[i];
             if (lmin[f] == lmax[f]) continue;
             wcand[ncand] = f;
-            if (a.splitter_random)
+            if (splitter_random)
                 wcbin[ncand] = lmin[f] + (int)philox_bounded(
                     wthr[f], (uint32_t)(lmax[f] - lmin[f]));
             ++ncand;
@@ -1213,7 +1225,7 @@ __device__ __forceinline__ void mid_wave_node(
     double best_s = -1.0e300;
     int bf = -1, bb = -1, bnl = 0;
 
-    if (a.splitter_random) {
+    if (splitter_random) {
         // ET: counts at each candidate's drawn bin — histogram-free.
         int cnt[WAVE_CANDS], cnt1[WAVE_CANDS];
         #pragma unroll
@@ -1466,6 +1478,8 @@ __global__ void mid_subtree_kernel(ForestDev a,
         auto block_node = [&](int ls, int le, int depth, int node,
                               bool dfs_mode) {
             const int n = le - ls;
+            const int max_features = a.j_mf[it.job];
+            const int splitter_random = a.j_rand[it.job];
             // global job-relative range (the RNG identity)
             const int gs = it.start + ls;
             const int ge = it.start + le;
@@ -1546,7 +1560,7 @@ __global__ void mid_subtree_kernel(ForestDev a,
                     int t = perm[i]; perm[i] = perm[j]; perm[j] = t;
                 }
                 int nc = 0;
-                for (int i = 0; i < F && nc < a.max_features; ++i) {
+                for (int i = 0; i < F && nc < max_features; ++i) {
                     int f = perm[i];
                     if (sh_bmin[f] != sh_bmax[f]) sh_cand[nc++] = f;
                 }
@@ -1585,7 +1599,7 @@ __global__ void mid_subtree_kernel(ForestDev a,
                 double best_s = -1.0;
                 int best_b = -1, best_nl = 0;
 
-                if (a.splitter_random) {
+                if (splitter_random) {
                     uint32_t tag = TAG_THRESH |
                                    ((uint32_t)(depth & 0xFF) << 8);
                     uint32_t u = philox_draw(tag, (uint32_t)gs,
@@ -1758,7 +1772,7 @@ __global__ void mid_subtree_kernel(ForestDev a,
             __syncthreads();
         };
 
-        if (a.wave_mid) {
+        if (a.wave_mid && a.j_mf[it.job] <= WAVE_CANDS) {
             // Wave-parallel rounds: up to HBLK/64 nodes in flight, each
             // built by one wave (no block barriers inside a node); a
             // single-node frontier — the top of every staged subtree —
@@ -1823,7 +1837,6 @@ __global__ void mid_subtree_kernel(ForestDev a,
         __syncthreads();
     }
 }
-}
 
 // ---------------------------------------------------------------------------
 // Wave-per-subtree builder for nodes with n <= SMALL_N (=64) samples.
@@ -1866,6 +1879,8 @@ __global__ void small_subtree_kernel(ForestDev a,
         const long nbase = a.j_node_off[it.job];
         const uint32_t key = (uint32_t)a.j_key[it.job];
         const int F = a.F;
+        const int max_features = a.j_mf[it.job];
+        const int splitter_random = a.j_rand[it.job];
 
         // lane -> sample (register-resident)
         uint32_t w[4] = {0, 0, 0, 0};
@@ -1925,7 +1940,7 @@ __global__ void small_subtree_kernel(ForestDev a,
             // lane 0's LDS writes are visible to the wave in program order
 
             uint32_t u_thresh = 0;
-            if (a.splitter_random) {
+            if (splitter_random) {
                 uint32_t ttag = TAG_THRESH |
                                 ((uint32_t)(depth & 0xFF) << 8);
                 u_thresh = philox_draw(ttag, (uint32_t)s, (uint32_t)e,
@@ -1937,7 +1952,7 @@ __global__ void small_subtree_kernel(ForestDev a,
             int best_f = -1, best_b = -1, best_nl = 0;
             int n_eval = 0;
 
-            for (int pi = 0; pi < F && n_eval < a.max_features; ++pi) {
+            for (int pi = 0; pi < F && n_eval < max_features; ++pi) {
                 const int f = perm[pi];
                 const int my_code = (int)((w[f >> 2] >> ((f & 3) * 8))
                                           & 0xFFu);
@@ -1953,7 +1968,7 @@ __global__ void small_subtree_kernel(ForestDev a,
                 if (cmin == cmax) continue;   // constant: not counted
                 ++n_eval;
 
-                if (a.splitter_random) {
+                if (splitter_random) {
                     const int b = cmin + (int)philox_bounded(
                         __shfl(u_thresh, f), (uint32_t)(cmax - cmin));
                     const unsigned long long lm =

@@ -32,10 +32,13 @@ static hipStream_t current_stream() {
 // ---------------------------------------------------------------------------
 // forest_fit
 // ---------------------------------------------------------------------------
-std::vector<at::Tensor> forest_fit(
+// Core fit over a batch of jobs with PER-JOB model spec (max_features,
+// splitter, bootstrap) — one call can build a whole balance group's
+// DT + RF + ET forests in a single level pipeline.
+static std::vector<at::Tensor> forest_fit_impl(
     at::Tensor codes, at::Tensor labels, at::Tensor j_row_off,
-    at::Tensor j_n, at::Tensor j_key, int64_t F, int64_t max_features,
-    bool bootstrap, bool splitter_random, int64_t seed) {
+    at::Tensor j_n, at::Tensor j_key, at::Tensor j_mf, at::Tensor j_rand,
+    at::Tensor j_boot, int64_t F, int64_t seed) {
     TORCH_CHECK(codes.is_cuda() && codes.dtype() == at::kByte &&
                 codes.size(1) == FPAD && codes.is_contiguous());
     TORCH_CHECK(labels.is_cuda() && labels.dtype() == at::kByte);
@@ -46,11 +49,20 @@ std::vector<at::Tensor> forest_fit(
     auto opts_i64 = codes.options().dtype(at::kLong);
     auto opts_f32 = codes.options().dtype(at::kFloat);
 
-    // j_n may arrive CPU-resident (preferred: the host needs its values
-    // for sizing, and a device tensor would force a blocking D2H here)
+    // Job arrays may arrive CPU-resident (preferred: the host needs
+    // j_n / j_rand values for sizing and kernel selection, and device
+    // tensors would force blocking D2H copies here)
     auto j_n_cpu = j_n.is_cuda() ? j_n.to(at::kCPU) : j_n;
     auto j_n_dev = j_n.is_cuda() ? j_n : j_n.to(codes.device());
+    auto j_rand_cpu = j_rand.is_cuda() ? j_rand.to(at::kCPU) : j_rand;
+    auto j_rand_dev = j_rand.is_cuda() ? j_rand : j_rand.to(codes.device());
+    auto j_mf_dev = j_mf.is_cuda() ? j_mf : j_mf.to(codes.device());
+    auto j_boot_dev = j_boot.is_cuda() ? j_boot : j_boot.to(codes.device());
     const int* jn = j_n_cpu.data_ptr<int>();
+    const uint8_t* jr = j_rand_cpu.data_ptr<uint8_t>();
+    bool any_rand = false, any_best = false;
+    for (int j = 0; j < J; ++j)
+        (jr[j] ? any_rand : any_best) = true;
     std::vector<long> sidx_off(J), node_off(J);
     long S = 0, Ntot = 0;
     bool has_wide = false;   // any node >= 2^16 samples needs the WIDE path
@@ -135,7 +147,8 @@ std::vector<at::Tensor> forest_fit(
         j_row_off.data_ptr<int>(), j_n_dev.data_ptr<int>(),
         j_sidx_off.data_ptr<long>(), j_key.data_ptr<int>(),
         node_alloc.data_ptr<int>(), sidx_a.data_ptr<int>(),
-        (WorkItem*)work_a.data_ptr(), bootstrap ? 1 : 0, (uint32_t)seed);
+        (WorkItem*)work_a.data_ptr(), j_boot_dev.data_ptr<uint8_t>(),
+        (uint32_t)seed);
     state.narrow(0, 0, 1).fill_((int)J);
 
     const int PINSZ = 64;
@@ -160,8 +173,8 @@ std::vector<at::Tensor> forest_fit(
     a.ncnt1 = ncnt1.data_ptr<float>();
     a.err_flag = err.data_ptr<int>();
     a.F = (int)F;
-    a.max_features = (int)max_features;
-    a.splitter_random = splitter_random ? 1 : 0;
+    a.j_mf = j_mf_dev.data_ptr<int>();
+    a.j_rand = j_rand_dev.data_ptr<uint8_t>();
     a.seed = (uint32_t)seed;
     a.work_cap = (int)work_cap;
     a.hist_pool0 = (uint32_t*)hist_pool0.data_ptr<int>();
@@ -174,10 +187,10 @@ std::vector<at::Tensor> forest_fit(
     a.mid_cap = (int)mid_cap;
     int* st = state.data_ptr<int>();
     a.pool_count = st + 1;   // kernel indexes [wp * 4]
-    // wave-parallel mid-subtree mode: RF / ET (max_features <= 4); the
-    // env var forces the block-serial DFS for same-box A/B runs
-    a.wave_mid = (max_features <= WAVE_CANDS &&
-                  !getenv("FLAKE16_NO_WAVE_MID")) ? 1 : 0;
+    // wave-parallel mid-subtree mode (per-item: jobs with
+    // max_features <= WAVE_CANDS); the env var forces the block-serial
+    // DFS for same-box A/B runs
+    a.wave_mid = getenv("FLAKE16_NO_WAVE_MID") ? 0 : 1;
 
     int GRID = 4096;
     if (const char* e = getenv("FLAKE16_FIT_GRID")) GRID = atoi(e);
@@ -196,18 +209,19 @@ std::vector<at::Tensor> forest_fit(
         a.nxt_count = st + nx * 4;
         a.small_count = st + nx * 4 + 2;
         a.mid_count = st + nx * 4 + 3;
-        if (splitter_random) {
-            et_split_kernel<<<GRID, HBLK, 0, stream>>>(a);
-        } else if (!has_wide && max_features < F &&
-                   getenv("FLAKE16_RF_CANDONLY")) {
-            // ablation variant: candidate-only RF histograms measured
-            // ~3% SLOWER than full histograms + subtraction pools
-            rf_cand_split_kernel<<<GRID, HBLK, 0, stream>>>(a);
-        } else {
-            hist_split_kernel<false><<<GRID, HBLK, 0, stream>>>(a);
-            if (has_wide)
-                hist_split_kernel<true><<<GRID, HBLK, 0, stream>>>(a);
+        if (any_best) {
+            if (!has_wide && getenv("FLAKE16_RF_CANDONLY")) {
+                // ablation variant: candidate-only RF histograms measured
+                // ~3% SLOWER than full histograms + subtraction pools
+                rf_cand_split_kernel<<<GRID, HBLK, 0, stream>>>(a);
+            } else {
+                hist_split_kernel<false><<<GRID, HBLK, 0, stream>>>(a);
+                if (has_wide)
+                    hist_split_kernel<true><<<GRID, HBLK, 0, stream>>>(a);
+            }
         }
+        if (any_rand)
+            et_split_kernel<<<GRID, HBLK, 0, stream>>>(a);
         mid_subtree_kernel<<<2048, HBLK, 0, stream>>>(a, a.sidx_nxt);
         small_subtree_kernel<<<2048, HBLK, 0, stream>>>(a, a.sidx_nxt);
         CHECK_HIP(hipMemcpyAsync(pinned_p + pinned_slot, st + nx * 4, 4,
@@ -267,6 +281,28 @@ std::vector<at::Tensor> forest_fit(
                 "forest_fit: work-queue capacity exceeded");
 
     return {nfeat, nsplit, nleft, ncnt0, ncnt1, j_node_off, node_alloc};
+}
+
+// Uniform-spec compatibility surface (single-model batches).
+std::vector<at::Tensor> forest_fit(
+    at::Tensor codes, at::Tensor labels, at::Tensor j_row_off,
+    at::Tensor j_n, at::Tensor j_key, int64_t F, int64_t max_features,
+    bool bootstrap, bool splitter_random, int64_t seed) {
+    const int J = j_n.size(0);
+    auto mf = at::full({J}, (int)max_features, at::kInt);
+    auto rnd = at::full({J}, splitter_random ? 1 : 0, at::kByte);
+    auto boot = at::full({J}, bootstrap ? 1 : 0, at::kByte);
+    return forest_fit_impl(codes, labels, j_row_off, j_n, j_key, mf, rnd,
+                           boot, F, seed);
+}
+
+// Mixed-spec batches (fused balance-group fits).
+std::vector<at::Tensor> forest_fit_multi(
+    at::Tensor codes, at::Tensor labels, at::Tensor j_row_off,
+    at::Tensor j_n, at::Tensor j_key, at::Tensor j_mf, at::Tensor j_rand,
+    at::Tensor j_boot, int64_t F, int64_t seed) {
+    return forest_fit_impl(codes, labels, j_row_off, j_n, j_key, j_mf,
+                           j_rand, j_boot, F, seed);
 }
 
 // ---------------------------------------------------------------------------
@@ -577,6 +613,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("forest_fit", &forest_fit,
           py::call_guard<py::gil_scoped_release>(),
           "Batched histogram-forest fit (gfx950)");
+    m.def("forest_fit_multi", &forest_fit_multi,
+          py::call_guard<py::gil_scoped_release>(),
+          "Mixed-model batched forest fit (per-job spec)");
     m.def("forest_predict_confusion", &forest_predict_confusion,
           py::call_guard<py::gil_scoped_release>(),
           "Ensemble predict + confusion accumulation");
