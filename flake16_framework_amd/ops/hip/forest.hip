@@ -127,6 +127,7 @@ __device__ __forceinline__ void route_child(const ForestDev& a,
 __global__ void forest_init_kernel(
     const int* __restrict__ j_row_off, const int* __restrict__ j_n,
     const long* __restrict__ j_sidx_off, const int* __restrict__ j_key,
+    const long* __restrict__ j_node_off, int* __restrict__ nfeat,
     int* __restrict__ node_alloc, int* __restrict__ sidx,
     WorkItem* __restrict__ work, const uint8_t* __restrict__ j_boot,
     uint32_t seed) {
@@ -150,6 +151,7 @@ __global__ void forest_init_kernel(
     }
     if (threadIdx.x == 0) {
         node_alloc[job] = 1;
+        nfeat[j_node_off[job]] = LEAF_SENTINEL;   // root starts as a leaf
         work[job] = {job, 0, 0, n, 0, -1};
     }
 }
@@ -421,6 +423,8 @@ __global__ void hist_split_kernel(ForestDev a) {
             sh_bestnL = bnl;
             if (bf >= 0) {
                 int l = atomicAdd(&a.node_alloc[it.job], 2);
+                a.nfeat[nbase + l] = LEAF_SENTINEL;       // children start
+                a.nfeat[nbase + l + 1] = LEAF_SENTINEL;   // as leaves
                 a.nfeat[nbase + it.node] = bf;
                 a.nsplit[nbase + it.node] = bb;
                 a.nleft[nbase + it.node] = l;
@@ -759,6 +763,8 @@ __global__ void et_split_kernel(ForestDev a) {
             sh_bestnL = bnl;
             if (bf >= 0) {
                 int l = atomicAdd(&a.node_alloc[it.job], 2);
+                a.nfeat[nbase + l] = LEAF_SENTINEL;       // children start
+                a.nfeat[nbase + l + 1] = LEAF_SENTINEL;   // as leaves
                 a.nfeat[nbase + it.node] = bf;
                 a.nsplit[nbase + it.node] = bb;
                 a.nleft[nbase + it.node] = l;
@@ -1033,6 +1039,8 @@ __global__ void rf_cand_split_kernel(ForestDev a) {
             sh_bestnL = bnl;
             if (bf >= 0) {
                 int l = atomicAdd(&a.node_alloc[it.job], 2);
+                a.nfeat[nbase + l] = LEAF_SENTINEL;       // children start
+                a.nfeat[nbase + l + 1] = LEAF_SENTINEL;   // as leaves
                 a.nfeat[nbase + it.node] = bf;
                 a.nsplit[nbase + it.node] = bb;
                 a.nleft[nbase + it.node] = l;
@@ -1348,6 +1356,8 @@ __device__ __forceinline__ void mid_wave_node(
     int lid = 0;
     if (lane == 0) {
         lid = atomicAdd(&a.node_alloc[it.job], 2);
+        a.nfeat[nbase + lid] = LEAF_SENTINEL;
+        a.nfeat[nbase + lid + 1] = LEAF_SENTINEL;
         a.nfeat[nbase + node] = bf;
         a.nsplit[nbase + node] = bb;
         a.nleft[nbase + node] = lid;
@@ -1675,6 +1685,8 @@ __global__ void mid_subtree_kernel(ForestDev a,
                 sh_bestnL = bnl;
                 if (bf >= 0) {
                     int l = atomicAdd(&a.node_alloc[it.job], 2);
+                    a.nfeat[nbase + l] = LEAF_SENTINEL;
+                    a.nfeat[nbase + l + 1] = LEAF_SENTINEL;
                     a.nfeat[nbase + node] = bf;
                     a.nsplit[nbase + node] = bb;
                     a.nleft[nbase + node] = l;
@@ -2045,6 +2057,8 @@ __global__ void small_subtree_kernel(ForestDev a,
             int lid = 0;
             if (lane == 0) {
                 lid = atomicAdd(&a.node_alloc[it.job], 2);
+                a.nfeat[nbase + lid] = LEAF_SENTINEL;
+                a.nfeat[nbase + lid + 1] = LEAF_SENTINEL;
                 a.nfeat[nbase + node] = best_f;
                 a.nsplit[nbase + node] = best_b;
                 a.nleft[nbase + node] = lid;

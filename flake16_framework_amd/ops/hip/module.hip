@@ -90,9 +90,12 @@ static std::vector<at::Tensor> forest_fit_impl(
     const long S_alloc = (S + GRAN - 1) / GRAN * GRAN;
     const long Ntot_alloc = (Ntot + GRAN - 1) / GRAN * GRAN;
 
-    auto nfeat = at::full({Ntot_alloc}, LEAF_SENTINEL, opts_i32);
-    // only nfeat needs initialization (LEAF sentinel); the other node
-    // fields are written for every reached node before any read
+    // No global sentinel fill (it was 1 GB/group at large N): every
+    // allocated node is written LEAF_SENTINEL at allocation time (root in
+    // forest_init_kernel, children at their parent's split), and the
+    // other node fields are written for every reached node before any
+    // read — so plain at::empty suffices for all five arrays.
+    auto nfeat = at::empty({Ntot_alloc}, opts_i32);
     auto nsplit = at::empty({Ntot_alloc}, opts_i32);
     auto nleft = at::empty({Ntot_alloc}, opts_i32);
     auto ncnt0 = at::empty({Ntot_alloc}, opts_f32);
@@ -104,7 +107,12 @@ static std::vector<at::Tensor> forest_fit_impl(
 
     // Histogram-subtraction pools (see forest.hip): sized for the worst
     // per-level allocation, 2 slots per splitting node >= HIST_SAVE_MIN.
-    int HIST_SAVE_MIN = 2048;
+    // Larger batches raise the subtraction threshold: with the pool
+    // clamped at 131072 slots, small thresholds exhaust it mid-band and
+    // the 16 KiB/slot store+load traffic overtakes the saved
+    // re-accumulation (measured at N=40k fused: 23.0 -> 15.3 s with
+    // 8192).
+    int HIST_SAVE_MIN = S < (1L << 24) ? 2048 : 8192;
     if (const char* e = getenv("FLAKE16_HIST_SAVE_MIN"))
         HIST_SAVE_MIN = atoi(e);
     // cap bounds pool memory at ~2 GB per parity (16 KiB per slot);
@@ -146,6 +154,7 @@ static std::vector<at::Tensor> forest_fit_impl(
     forest_init_kernel<<<J, HBLK, 0, stream>>>(
         j_row_off.data_ptr<int>(), j_n_dev.data_ptr<int>(),
         j_sidx_off.data_ptr<long>(), j_key.data_ptr<int>(),
+        j_node_off.data_ptr<long>(), nfeat.data_ptr<int>(),
         node_alloc.data_ptr<int>(), sidx_a.data_ptr<int>(),
         (WorkItem*)work_a.data_ptr(), j_boot_dev.data_ptr<uint8_t>(),
         (uint32_t)seed);
