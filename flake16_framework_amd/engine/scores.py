@@ -190,11 +190,15 @@ def _run_scores_hip(all_cells, tests, tests_file, seed, progress,
     # stream while other workers proceed with other groups.
     context = SweepContext(tests=tests, tests_file=tests_file, seed=seed)
 
-    groups = {}
-    for cell_idx, config_keys in all_cells:
-        g = balance_group_index(config_keys)
-        groups.setdefault(g, []).append((config_keys, cell_idx))
-    tasks = [groups[g] for g in sorted(groups)]
+    if os.environ.get("FLAKE16_NO_GROUP_FUSE"):
+        tasks = [[(config_keys, cell_idx)]
+                 for cell_idx, config_keys in all_cells]
+    else:
+        groups = {}
+        for cell_idx, config_keys in all_cells:
+            g = balance_group_index(config_keys)
+            groups.setdefault(g, []).append((config_keys, cell_idx))
+        tasks = [groups[g] for g in sorted(groups)]
 
     out = {}
     lock = threading.Lock()
