@@ -142,12 +142,16 @@ def write_table(table_file, tab, rowcol=True, cellfn=cellfn_default):
 # ---------------------------------------------------------------------------
 
 def load_subjects(tests, subjects_file=None):
-    """[(proj, repo)] — from subjects.txt when available (reference row
-    order and 'owner/repo' first column), else the tests.json keys."""
+    """[(proj, repo)] — from subjects.txt when it describes THIS dataset
+    (reference row order and 'owner/repo' first column), else the
+    tests.json keys.  A synthetic tests.json has proj00..proj25 keys, so
+    the real study subject list must not be applied to it."""
     path = subjects_file or SUBJECTS_FILE
     if os.path.exists(path):
         from ..orchestrate.runner import read_subjects
-        return [(s.proj, s.repo) for s in read_subjects(path)]
+        subjects = [(s.proj, s.repo) for s in read_subjects(path)]
+        if all(proj in tests for proj, _ in subjects):
+            return subjects
     return [(proj, proj) for proj in tests.keys()]
 
 
