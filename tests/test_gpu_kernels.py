@@ -284,6 +284,31 @@ class TestCellE2E:
             assert got[2] == ref[2], keys
             assert got[3] == ref[3], keys
 
+    def test_fused_group_equals_per_cell(self, ops):
+        """The production sweep path fuses a balance group's 3 model
+        cells into one mixed-model fit — results must equal per-cell
+        evaluation bit-for-bit (same Philox keys per job)."""
+        from flake16_framework_amd.configgrid import iter_config_keys
+        from flake16_framework_amd.dataset.synthetic import (
+            make_synthetic_tests,
+        )
+        from flake16_framework_amd.engine.hip_cell import SweepContext
+
+        tests = make_synthetic_tests(n_tests=400, seed=2)
+        all_keys = list(iter_config_keys())
+        group = [k for k in all_keys
+                 if k[:4] == ("NOD", "Flake16", "None", "SMOTE")]
+        assert len(group) == 3
+        cells = [(k, all_keys.index(k)) for k in group]
+
+        ctx = SweepContext(tests=tests)
+        fused = ctx.evaluate_group(cells)
+        for cell in cells:
+            single = SweepContext(tests=tests).evaluate_group([cell])
+            keys = cell[0]
+            assert fused[keys][2] == single[keys][2], keys
+            assert fused[keys][3] == single[keys][3], keys
+
     def test_preproc_cells_metric_close(self, ops):
         from flake16_framework_amd.configgrid import iter_config_keys
         from flake16_framework_amd.dataset.synthetic import (
