@@ -1809,6 +1809,15 @@ __global__ void mid_subtree_kernel(ForestDev a,
                 if (sh_take == 1) {
                     const MidFrame fr = wframes[0];
                     block_node(fr.s, fr.e, fr.depth, fr.node, false);
+                } else if (sh_take == 2 &&
+                           (wframes[0].e - wframes[0].s)
+                               + (wframes[1].e - wframes[1].s) > 1024) {
+                    // two big frontier nodes: sequential block-wide
+                    // processing beats two waves with two idle
+                    for (int x = 0; x < 2; ++x) {
+                        const MidFrame fr = wframes[x];
+                        block_node(fr.s, fr.e, fr.depth, fr.node, false);
+                    }
                 } else if (wave < sh_take) {
                     const MidFrame fr = wframes[wave];
                     mid_wave_node(a, it, fr.s, fr.e, fr.depth, fr.node,
