@@ -1142,11 +1142,24 @@ struct MidFrame {
 // identical to the block-wide path — trees are bit-identical.  Requires
 // max_features <= WAVE_CANDS (RF / ET; DT keeps the block path).
 // whist: this wave's WAVE_CANDS*256-word LDS histogram region (RF only).
+template <bool STAGED>
 __device__ __forceinline__ void mid_wave_node(
     const ForestDev& a, const WorkItem& it, int ls, int le, int depth,
-    int node, const uint4* m_codes, const uint8_t* m_lab, uint16_t* m_idx,
+    int node, const uint4* m_codes, const uint8_t* m_lab,
+    const int* m_rows, uint16_t* m_idx,
     uint16_t* m_idx2, uint32_t* whist, int* wperm, uint32_t* wthr,
     int* wcand, int* wcbin, MidFrame* stack, int* sh_count) {
+    // STAGED reads the LDS-staged code rows / labels; the unstaged
+    // variant reads them from L1/L2 through the m_rows map (half the
+    // LDS per block -> double the resident blocks).
+    auto code_at = [&](int o) -> uint4 {
+        if (STAGED) return m_codes[o];
+        return *reinterpret_cast<const uint4*>(
+            a.codes + (size_t)m_rows[o] * FPAD);
+    };
+    auto lab_at = [&](int o) -> int {
+        return STAGED ? (int)m_lab[o] : (int)a.labels[m_rows[o]];
+    };
     const int lane = threadIdx.x & 63;
     const int n = le - ls;
     const long nbase = a.j_node_off[it.job];
@@ -1164,9 +1177,9 @@ __device__ __forceinline__ void mid_wave_node(
     int lc1 = 0;
     for (int i = ls + lane; i < le; i += 64) {
         const int o = m_idx[i];
-        const uint4 cw = m_codes[o];
+        const uint4 cw = code_at(o);
         const uint32_t w[4] = {cw.x, cw.y, cw.z, cw.w};
-        lc1 += m_lab[o];
+        lc1 += lab_at(o);
         #pragma unroll
         for (int f = 0; f < FPAD; ++f) {
             const int b = (int)((w[f >> 2] >> ((f & 3) * 8)) & 0xFFu);
@@ -1240,9 +1253,9 @@ __device__ __forceinline__ void mid_wave_node(
         for (int ci = 0; ci < WAVE_CANDS; ++ci) { cnt[ci] = 0; cnt1[ci] = 0; }
         for (int i = ls + lane; i < le; i += 64) {
             const int o = m_idx[i];
-            const uint4 cw = m_codes[o];
+            const uint4 cw = code_at(o);
             const uint32_t w[4] = {cw.x, cw.y, cw.z, cw.w};
-            const int lab = m_lab[o];
+            const int lab = lab_at(o);
             #pragma unroll
             for (int ci = 0; ci < WAVE_CANDS; ++ci) {
                 if (ci >= ncand) break;
@@ -1281,9 +1294,9 @@ __device__ __forceinline__ void mid_wave_node(
         for (int i = lane; i < ncand * 256; i += 64) whist[i] = 0;
         for (int i = ls + lane; i < le; i += 64) {
             const int o = m_idx[i];
-            const uint4 cw = m_codes[o];
+            const uint4 cw = code_at(o);
             const uint32_t w[4] = {cw.x, cw.y, cw.z, cw.w};
-            const uint32_t inc = 1u | ((uint32_t)m_lab[o] << 16);
+            const uint32_t inc = 1u | ((uint32_t)lab_at(o) << 16);
             #pragma unroll
             for (int ci = 0; ci < WAVE_CANDS; ++ci) {
                 if (ci >= ncand) break;
@@ -1373,7 +1386,7 @@ __device__ __forceinline__ void mid_wave_node(
         int o = 0, flag = 0;
         if (valid) {
             o = m_idx[i];
-            const uint4 cw = m_codes[o];
+            const uint4 cw = code_at(o);
             const uint32_t wsel = (&cw.x)[bf >> 2];
             flag = (int)(((wsel >> ((bf & 3) * 8)) & 0xFFu)
                          <= (uint32_t)bb);
@@ -1430,12 +1443,13 @@ __device__ __forceinline__ void mid_wave_node(
     }
 }
 
+template <bool STAGED>
 __launch_bounds__(HBLK)
 __global__ void mid_subtree_kernel(ForestDev a,
                                    const int* __restrict__ sidx_level) {
-    __shared__ uint4 m_codes[MID_N];          // 16 KiB staged code rows
+    __shared__ uint4 m_codes[STAGED ? MID_N : 1];   // 32 KiB staged rows
     __shared__ int m_rows[MID_N];             // 4 KiB original global rows
-    __shared__ uint8_t m_lab[MID_N];          // 1 KiB labels
+    __shared__ uint8_t m_lab[STAGED ? MID_N : 1];   // 2 KiB labels
     __shared__ uint16_t m_idx[MID_N];         // 2 KiB sample ordering
     __shared__ uint16_t m_idx2[MID_N];        // 2 KiB partition scratch
     __shared__ uint32_t hist[FPAD * 256];     // 16 KiB packed histogram
@@ -1475,11 +1489,22 @@ __global__ void mid_subtree_kernel(ForestDev a,
         for (int i = tid; i < n0; i += HBLK) {
             const int row = sidx_level[sbase + it.start + i];
             m_rows[i] = row;
-            m_codes[i] = *reinterpret_cast<const uint4*>(
-                a.codes + (size_t)row * FPAD);
-            m_lab[i] = a.labels[row];
+            if (STAGED) {
+                m_codes[i] = *reinterpret_cast<const uint4*>(
+                    a.codes + (size_t)row * FPAD);
+                m_lab[i] = a.labels[row];
+            }
             m_idx[i] = (uint16_t)i;
         }
+
+        auto code_at = [&](int o) -> uint4 {
+            if (STAGED) return m_codes[o];
+            return *reinterpret_cast<const uint4*>(
+                a.codes + (size_t)m_rows[o] * FPAD);
+        };
+        auto lab_at = [&](int o) -> int {
+            return STAGED ? (int)m_lab[o] : (int)a.labels[m_rows[o]];
+        };
 
         // Per-node block-wide builder (histogram, split, partition over
         // the staged window) shared by the DFS path and the wave mode's
@@ -1500,9 +1525,9 @@ __global__ void mid_subtree_kernel(ForestDev a,
             __syncthreads();
             for (int i = ls + tid; i < le; i += HBLK) {
                 const int o = m_idx[i];
-                const uint4 cw = m_codes[o];
+                const uint4 cw = code_at(o);
                 const uint32_t w[4] = {cw.x, cw.y, cw.z, cw.w};
-                const uint32_t inc = 1u | ((uint32_t)m_lab[o] << 16);
+                const uint32_t inc = 1u | ((uint32_t)lab_at(o) << 16);
                 for (int f = 0; f < F; ++f) {
                     uint32_t b = (w[f >> 2] >> ((f & 3) * 8)) & 0xFFu;
                     atomicAdd(&hist[f * 256 + b], inc);
@@ -1712,7 +1737,7 @@ __global__ void mid_subtree_kernel(ForestDev a,
                 int o = 0, flag = 0;
                 if (valid) {
                     o = m_idx[i];
-                    const uint4 cw = m_codes[o];
+                    const uint4 cw = code_at(o);
                     const uint32_t wsel = (&cw.x)[bf >> 2];
                     flag = (int)(((wsel >> ((bf & 3) * 8)) & 0xFFu)
                                  <= (uint32_t)bb);
@@ -1820,12 +1845,13 @@ __global__ void mid_subtree_kernel(ForestDev a,
                     }
                 } else if (wave < sh_take) {
                     const MidFrame fr = wframes[wave];
-                    mid_wave_node(a, it, fr.s, fr.e, fr.depth, fr.node,
-                                  m_codes, m_lab, m_idx, m_idx2,
-                                  hist + wave * (WAVE_CANDS * 256),
-                                  wperm_ws[wave], wthr_ws[wave],
-                                  wcand_ws[wave], wcbin_ws[wave],
-                                  stack, &sh_count);
+                    mid_wave_node<STAGED>(
+                        a, it, fr.s, fr.e, fr.depth, fr.node,
+                        m_codes, m_lab, m_rows, m_idx, m_idx2,
+                        hist + wave * (WAVE_CANDS * 256),
+                        wperm_ws[wave], wthr_ws[wave],
+                        wcand_ws[wave], wcbin_ws[wave],
+                        stack, &sh_count);
                 }
             }
             __syncthreads();

@@ -203,6 +203,10 @@ static std::vector<at::Tensor> forest_fit_impl(
 
     int GRID = 4096;
     if (const char* e = getenv("FLAKE16_FIT_GRID")) GRID = atoi(e);
+    // STAGED keeps code rows in LDS (2 blocks/CU); the unstaged variant
+    // reads them through L1/L2 and doubles the resident blocks — A/B via
+    // FLAKE16_MID_UNSTAGED=1.
+    const bool mid_staged = !getenv("FLAKE16_MID_UNSTAGED");
 
     // One level's dispatches: clear next-parity counters, split kernel(s),
     // mid- and small-subtree drains, next-level count -> pinned slot.
@@ -231,7 +235,12 @@ static std::vector<at::Tensor> forest_fit_impl(
         }
         if (any_rand)
             et_split_kernel<<<GRID, HBLK, 0, stream>>>(a);
-        mid_subtree_kernel<<<2048, HBLK, 0, stream>>>(a, a.sidx_nxt);
+        if (mid_staged)
+            mid_subtree_kernel<true><<<2048, HBLK, 0, stream>>>(
+                a, a.sidx_nxt);
+        else
+            mid_subtree_kernel<false><<<2048, HBLK, 0, stream>>>(
+                a, a.sidx_nxt);
         small_subtree_kernel<<<2048, HBLK, 0, stream>>>(a, a.sidx_nxt);
         CHECK_HIP(hipMemcpyAsync(pinned_p + pinned_slot, st + nx * 4, 4,
                                  hipMemcpyDeviceToHost, stream));
