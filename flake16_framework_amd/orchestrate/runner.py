@@ -80,26 +80,50 @@ def read_subjects(subjects_file=SUBJECTS_FILE):
 # Subject environment provisioning (the `setup` stage)
 # ---------------------------------------------------------------------------
 
-def provision_subject(subject, subjects_dir=SUBJECTS_DIR):
+def create_venv(venv_dir, run=sp.run):
+    """virtualenv when available (the reference's tool; the runner image
+    installs it), `python -m venv` otherwise, degrading to --without-pip
+    where ensurepip has no bundled wheels — every variant yields the
+    bin/ + site-packages layout the rest relies on."""
+    import shutil
+    if shutil.which("virtualenv"):
+        run(["virtualenv", venv_dir], check=True)
+        return
+    try:
+        run([sys.executable, "-m", "venv", venv_dir], check=True)
+    except sp.CalledProcessError:
+        shutil.rmtree(venv_dir, ignore_errors=True)
+        run([sys.executable, "-m", "venv", "--without-pip", venv_dir],
+            check=True)
+
+
+def provision_subject(subject, subjects_dir=SUBJECTS_DIR, run=sp.run,
+                      pip_run=None):
     """Create the subject's venv, clone at the pinned SHA, install the
-    pinned requirements + the project, and expose the collectors."""
+    pinned requirements + the project, and expose the collectors.
+
+    run / pip_run: injectable process runners (pip_run defaults to run);
+    FLAKE16_GIT_BASE overrides the clone base URL (local mirrors, tests).
+    """
+    pip_run = pip_run or run
     root = os.path.join(subjects_dir, subject.proj)
     checkout = os.path.join(root, subject.proj)
     venv_dir = os.path.join(root, "venv")
     pins = os.path.join(root, "requirements.txt")
+    git_base = os.environ.get("FLAKE16_GIT_BASE", "https://github.com/")
 
     env = os.environ.copy()
     env["PATH"] = os.path.join(venv_dir, "bin") + ":" + env["PATH"]
 
-    sp.run(["virtualenv", venv_dir], check=True)
-    sp.run(["git", "clone", f"https://github.com/{subject.repo}", checkout],
-           check=True)
-    sp.run(["git", "reset", "--hard", subject.sha], cwd=checkout, check=True)
+    create_venv(venv_dir, run=run)
+    run(["git", "clone", git_base + subject.repo, checkout], check=True)
+    run(["git", "reset", "--hard", subject.sha], cwd=checkout, check=True)
 
-    sp.run([*PIP_INSTALL, PIP_VERSION], env=env, check=True)
-    sp.run([*PIP_INSTALL, "-r", pins], env=env, check=True)
-    sp.run([*PIP_INSTALL, "-e", os.path.join(checkout, subject.package_dir)],
-           env=env, check=True)
+    pip_run([*PIP_INSTALL, PIP_VERSION], env=env, check=True)
+    pip_run([*PIP_INSTALL, "-r", pins], env=env, check=True)
+    pip_run([*PIP_INSTALL, "-e",
+             os.path.join(checkout, subject.package_dir)],
+            env=env, check=True)
     install_plugins(venv_dir)
 
 

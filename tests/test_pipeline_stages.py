@@ -66,6 +66,65 @@ class TestOrchestrate:
         assert "-p" in final and "no:randomly" in final
         assert calls[1][1]["timeout"] == 7200
 
+    def test_provision_subject_executes(self, tmp_path, monkeypatch):
+        """Real execution of the setup stage minus the pip installs
+        (there is no package index offline): venv created, subject
+        cloned at the pinned SHA from a local mirror, collection
+        plugins exposed via .pth, pip argv recorded in order."""
+        import subprocess
+        from flake16_framework_amd.orchestrate.runner import (
+            FRAMEWORK_ROOT, Subject, provision_subject,
+        )
+
+        # local "upstream" repo with two commits; pin the FIRST one
+        upstream = tmp_path / "mirror" / "alice" / "proj-x"
+        upstream.mkdir(parents=True)
+        git = ["git", "-C", str(upstream), "-c", "user.email=t@t",
+               "-c", "user.name=t"]
+        subprocess.run([*git[:3], "init", "-q"], check=True)
+        (upstream / "setup.py").write_text("#\n")
+        subprocess.run([*git, "add", "-A"], check=True)
+        subprocess.run([*git, "commit", "-q", "-m", "c1"], check=True)
+        sha = subprocess.run([*git, "rev-parse", "HEAD"],
+                             capture_output=True, text=True,
+                             check=True).stdout.strip()
+        (upstream / "extra.txt").write_text("later\n")
+        subprocess.run([*git, "add", "-A"], check=True)
+        subprocess.run([*git, "commit", "-q", "-m", "c2"], check=True)
+
+        subjects_dir = tmp_path / "subjects"
+        (subjects_dir / "proj-x").mkdir(parents=True)
+        (subjects_dir / "proj-x" / "requirements.txt").write_text("")
+
+        pip_calls = []
+
+        def record_pip(argv, **kwargs):
+            pip_calls.append(argv)
+
+        monkeypatch.setenv("FLAKE16_GIT_BASE",
+                           str(tmp_path / "mirror") + "/")
+        subject = Subject("proj-x", "alice/proj-x", sha, ".", ("pytest",))
+        provision_subject(subject, subjects_dir=str(subjects_dir),
+                          pip_run=record_pip)
+
+        venv = subjects_dir / "proj-x" / "venv"
+        assert (venv / "bin" / "python").exists()
+        checkout = subjects_dir / "proj-x" / "proj-x"
+        head = subprocess.run(["git", "-C", str(checkout), "rev-parse",
+                               "HEAD"], capture_output=True, text=True,
+                              check=True).stdout.strip()
+        assert head == sha                      # pinned, not the tip
+        assert not (checkout / "extra.txt").exists()
+
+        import glob
+        pths = glob.glob(str(venv / "lib" / "python*" / "site-packages"
+                             / "flake16_framework_amd.pth"))
+        assert pths and open(pths[0]).read().strip() == FRAMEWORK_ROOT
+
+        assert [c[-1] for c in pip_calls][0].startswith("pip==")
+        assert pip_calls[1][-2] == "-r"
+        assert pip_calls[2][-2] == "-e"
+
     def test_run_log_resume(self, tmp_path):
         from flake16_framework_amd.orchestrate.runner import read_log
         log = tmp_path / "log.txt"
