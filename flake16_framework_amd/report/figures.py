@@ -323,6 +323,11 @@ def write_figures(tests_file=TESTS_FILE, scores_file=SCORES_FILE,
     write_table(dest("od-top.tex"), tab_od_top)
 
     for name, (orig_keys, ext_keys) in COMPARISON_CONFIGS.items():
+        if orig_keys not in scores or ext_keys not in scores:
+            # partial sweep (e.g. --cells subsets): emit an empty table
+            # instead of failing the whole stage
+            TexTable().write(dest(f"{name}-comp.tex"))
+            continue
         comparison_table(scores, orig_keys, ext_keys).write(
             dest(f"{name}-comp.tex"))
 
