@@ -208,6 +208,13 @@ static std::vector<at::Tensor> forest_fit_impl(
     // FLAKE16_MID_UNSTAGED=1.
     const bool mid_staged = !getenv("FLAKE16_MID_UNSTAGED");
 
+    // Adaptive level grids: the deep tail of the band (skewed-split
+    // chains) has a handful of items per level, where scheduling the
+    // full 4096/2048 empty-block grids is the dominant cost; once the
+    // synced queue count is tiny the next chunk launches small grids
+    // (grid-stride loops keep any count correct regardless).
+    int lv_grid = GRID, sub_grid = 2048;
+
     // One level's dispatches: clear next-parity counters, split kernel(s),
     // mid- and small-subtree drains, next-level count -> pinned slot.
     auto level_ops = [&](int cur_par, int pinned_slot) {
@@ -226,22 +233,22 @@ static std::vector<at::Tensor> forest_fit_impl(
             if (!has_wide && getenv("FLAKE16_RF_CANDONLY")) {
                 // ablation variant: candidate-only RF histograms measured
                 // ~3% SLOWER than full histograms + subtraction pools
-                rf_cand_split_kernel<<<GRID, HBLK, 0, stream>>>(a);
+                rf_cand_split_kernel<<<lv_grid, HBLK, 0, stream>>>(a);
             } else {
-                hist_split_kernel<false><<<GRID, HBLK, 0, stream>>>(a);
+                hist_split_kernel<false><<<lv_grid, HBLK, 0, stream>>>(a);
                 if (has_wide)
-                    hist_split_kernel<true><<<GRID, HBLK, 0, stream>>>(a);
+                    hist_split_kernel<true><<<lv_grid, HBLK, 0, stream>>>(a);
             }
         }
         if (any_rand)
-            et_split_kernel<<<GRID, HBLK, 0, stream>>>(a);
+            et_split_kernel<<<lv_grid, HBLK, 0, stream>>>(a);
         if (mid_staged)
-            mid_subtree_kernel<true><<<2048, HBLK, 0, stream>>>(
+            mid_subtree_kernel<true><<<sub_grid, HBLK, 0, stream>>>(
                 a, a.sidx_nxt);
         else
-            mid_subtree_kernel<false><<<2048, HBLK, 0, stream>>>(
+            mid_subtree_kernel<false><<<sub_grid, HBLK, 0, stream>>>(
                 a, a.sidx_nxt);
-        small_subtree_kernel<<<2048, HBLK, 0, stream>>>(a, a.sidx_nxt);
+        small_subtree_kernel<<<sub_grid, HBLK, 0, stream>>>(a, a.sidx_nxt);
         CHECK_HIP(hipMemcpyAsync(pinned_p + pinned_slot, st + nx * 4, 4,
                                  hipMemcpyDeviceToHost, stream));
     };
@@ -291,6 +298,12 @@ static std::vector<at::Tensor> forest_fit_impl(
             CHECK_HIP(hipStreamSynchronize(stream));
             for (int c = 0; c < CHUNK; ++c)
                 if (pinned_p[c] == 0) { done = true; break; }
+            // shrink the next chunk's grids when the band has thinned
+            // (items can at most double per level: 16 items now bound
+            // the whole next chunk well under a 512-block grid)
+            const int last_cnt = pinned_p[CHUNK - 1];
+            lv_grid = last_cnt <= 16 ? 512 : GRID;
+            sub_grid = last_cnt <= 16 ? 512 : 2048;
             TORCH_CHECK(lev < 8192, "forest_fit: depth limit exceeded");
         }
     }
