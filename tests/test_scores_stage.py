@@ -151,6 +151,62 @@ class TestDistributedGloo:
                  for r in range(8)]
         assert max(loads) / min(loads) < 1.3
 
+    def test_shard_never_splits_balance_groups(self):
+        """The fused mixed-model fit assumes a rank owns WHOLE balance
+        groups (the 3 model cells share balanced folds and one fit)."""
+        from flake16_framework_amd.configgrid import balance_group_index
+        from flake16_framework_amd.parallel.comm import shard_cells
+        for world in (2, 4, 8):
+            owner = {}
+            for r in range(world):
+                for c in shard_cells(world, r):
+                    g = balance_group_index(ALL_KEYS[c])
+                    assert owner.setdefault(g, r) == r, (world, g)
+
+    def test_checkpoint_reshard_drops_foreign_cells(self, tmp_path,
+                                                    monkeypatch):
+        """A resume under a different world size must not merge cells the
+        current shard does not own (they would double-count through the
+        all-reduce SUM)."""
+        import json
+        import flake16_framework_amd.engine.scores as scores_mod
+        from flake16_framework_amd.engine.scores import write_scores
+        from flake16_framework_amd.parallel import comm
+
+        tests = _small_tests(400, seed=2)
+        monkeypatch.chdir(tmp_path)
+        with open(tmp_path / "tests.json", "w") as fd:
+            json.dump(tests, fd)
+
+        dt = [i for i, k in enumerate(ALL_KEYS)
+              if k[4] == "Decision Tree"][:4]
+        ckpt = str(tmp_path / "ck")
+
+        # "old world" run owned 4 cells and checkpointed them all
+        monkeypatch.setattr(comm, "shard_cells",
+                            lambda world, rank, n_cells=None: dt)
+        write_scores(tests_file=str(tmp_path / "tests.json"),
+                     scores_file=str(tmp_path / "s.pkl"),
+                     backend="ref", checkpoint=ckpt)
+
+        # "new world": this rank owns only the first two of them — the
+        # other two checkpointed cells must be ignored, not merged
+        monkeypatch.setattr(comm, "shard_cells",
+                            lambda world, rank, n_cells=None: dt[:2])
+        evaluated = []
+        real_eval = scores_mod.evaluate_cell_ref
+
+        def spy(keys, ci, **kw):
+            evaluated.append(ci)
+            return real_eval(keys, ci, **kw)
+
+        monkeypatch.setattr(scores_mod, "evaluate_cell_ref", spy)
+        r2 = write_scores(tests_file=str(tmp_path / "tests.json"),
+                         scores_file=str(tmp_path / "s.pkl"),
+                         backend="ref", checkpoint=ckpt)
+        assert not evaluated                 # both owned cells checkpointed
+        assert set(r2) == {ALL_KEYS[c] for c in dt[:2]}
+
 
 class TestCheckpointAndTrace:
     def test_checkpoint_resume_skips_done_cells(self, tmp_path, monkeypatch):
