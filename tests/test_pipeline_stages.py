@@ -365,6 +365,32 @@ class TestReportHelpers:
         assert "\\rowcolor{gray!20}" in text  # zebra on odd rows
         assert "a & 1.00 \\\\" in text
 
+    def test_write_table_exact_bytes(self, tmp_path):
+        """Byte contract of the table writer (the rewrite was verified
+        byte-identical to the round-1 implementation on all eight
+        artifacts; this pins the core rendering rules)."""
+        from flake16_framework_amd.report.figures import write_table
+        p = tmp_path / "t.tex"
+        write_table(str(p), [[["a", 1.0, 0], ["b", -0.5, 3]], [["T", 2]]])
+        assert p.read_text() == (
+            "a & 1.00 & - \\\\\n"
+            "\\rowcolor{gray!20}\n"
+            "b & -0.50 & 3 \\\\\n"
+            "\\midrule\n"
+            "T & 2 \\\\\n")
+
+    def test_req_runs_plot_exact_bytes(self, tmp_path):
+        from flake16_framework_amd.report.figures import (
+            write_req_runs_plot,
+        )
+        p = tmp_path / "rr.tex"
+        write_req_runs_plot({100: 1}, {}, str(p))
+        text = p.read_text()
+        assert text.startswith(
+            "\\addplot[mark=x,only marks] coordinates {(100,1.0) (200,1.0)")
+        assert text.endswith("\\addlegendentry{OD}")   # no trailing newline
+        assert "(2500,0.0)" in text                    # empty OD histogram
+
     def test_req_runs_plot_coords_normalized(self):
         from flake16_framework_amd.report.figures import (
             get_req_runs_plot_coords,
