@@ -271,6 +271,20 @@ __global__ void treeshap_paths_kernel(
         double pz = 1.0, po = 1.0;
         int pi = -1;
 
+        // feature -> path-index map, byte-packed in two registers (every
+        // feature appears at most once on the path), replacing an O(l)
+        // scratch scan per step
+        uint64_t fmap0 = ~0ULL, fmap1 = ~0ULL;
+        auto fm_get = [&](int f) -> int {
+            const uint64_t w = f < 8 ? fmap0 : fmap1;
+            return (int)((w >> ((f & 7) * 8)) & 0xFF);
+        };
+        auto fm_set = [&](int f, int v) {
+            uint64_t& w = f < 8 ? fmap0 : fmap1;
+            const int s = (f & 7) * 8;
+            w = (w & ~(0xFFULL << s)) | ((uint64_t)(v & 0xFF) << s);
+        };
+
         for (int step = 0; step < plen; ++step) {
             const int node = path_nodes[p0 + step];
 
@@ -280,18 +294,25 @@ __global__ void treeshap_paths_kernel(
             // the (pz, po) carried into this extend; so: extend first,
             // then (for internal nodes) prepare the child's (pz, po).
             l = shap_extend(m, l, pz, po, pi);
+            if (pi >= 0) fm_set(pi, l - 1);
 
             const int f = nfeat[nb + node];
             if (f < 0) break;   // the leaf itself: extended, done
 
+            int k = fm_get(f);
+            if (k == 0xFF) k = -1;
             double iz = 1.0, io = 1.0;
-            int k = -1;
-            for (int i = 0; i < l; ++i)
-                if (m[i].d == f) { k = i; break; }
             if (k >= 0) {
                 iz = m[k].z;
                 io = m[k].o;
                 l = shap_unwind(m, l, k);
+                #pragma unroll
+                for (int ff = 0; ff < 16; ++ff) {
+                    const int p = fm_get(ff);
+                    if (p == 0xFF) continue;
+                    if (ff == f) fm_set(ff, 0xFF);
+                    else if (p > k) fm_set(ff, p - 1);
+                }
             }
 
             const int child = path_nodes[p0 + step + 1];

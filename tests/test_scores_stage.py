@@ -151,6 +151,21 @@ class TestDistributedGloo:
                  for r in range(8)]
         assert max(loads) / min(loads) < 1.3
 
+    def test_ref_process_pool_matches_serial(self):
+        """run_scores(processes=2) — the reference's Pool execution model
+        — must produce the serial results."""
+        from flake16_framework_amd.engine.scores import run_scores
+        tests = _small_tests(300, seed=4)
+        cells = [i for i, k in enumerate(ALL_KEYS)
+                 if k[4] == "Decision Tree" and k[3] == "None"][:3]
+        serial = run_scores(tests=tests, backend="ref", cells=cells)
+        pooled = run_scores(tests=tests, backend="ref", cells=cells,
+                            processes=2)
+        assert serial.keys() == pooled.keys()
+        for k in serial:
+            assert pooled[k][2] == serial[k][2]
+            assert pooled[k][3] == serial[k][3]
+
     def test_shard_never_splits_balance_groups(self):
         """The fused mixed-model fit assumes a rank owns WHOLE balance
         groups (the 3 model cells share balanced folds and one fit)."""
