@@ -88,6 +88,39 @@ class TestLabeling:
                         N_RUNS_SMALL) == (0, None)
 
 
+class TestNumbits:
+    """Known-answer vectors for coverage.py's numbits packing (bit i of
+    byte b = line b*8+i) — pins the format both the collector's encoder
+    and the collation decoder implement (coverage.py itself is not
+    installable here)."""
+
+    def test_encode_known_answer(self):
+        from flake16_framework_amd.collect.testinspect import (
+            nums_to_numbits,
+        )
+        assert nums_to_numbits({1, 2, 9}) == b"\x06\x02"
+        assert nums_to_numbits({0}) == b"\x01"
+        assert nums_to_numbits(set()) == b""
+        assert nums_to_numbits({15}) == b"\x00\x80"
+
+    def test_decode_known_answer(self):
+        from flake16_framework_amd.dataset.collate import _numbits_to_nums
+        assert _numbits_to_nums(b"\x06\x02") == [1, 2, 9]
+        assert _numbits_to_nums(b"\x00\x80") == [15]
+        assert _numbits_to_nums(b"") == []
+
+    def test_roundtrip(self):
+        import random
+        from flake16_framework_amd.collect.testinspect import (
+            nums_to_numbits,
+        )
+        from flake16_framework_amd.dataset.collate import _numbits_to_nums
+        rng = random.Random(5)
+        for _ in range(20):
+            nums = {rng.randrange(1, 400) for _ in range(rng.randrange(40))}
+            assert set(_numbits_to_nums(nums_to_numbits(nums))) == nums
+
+
 class TestCoverageFeatures:
     def test_basic_counts(self):
         cov = {"src/a.py": {1, 2, 3}, "tests/t.py": {5, 6}}
