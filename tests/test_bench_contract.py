@@ -61,7 +61,7 @@ def test_self_launch_two_ranks():
 @pytest.mark.timeout(600)
 def test_shap_stage_contract():
     out = _run_bench("--stage", "shap", "--backend", "ref", "--n-tests",
-                     "250", "--steps", "1", "--warmup", "0")
+                     "120", "--steps", "1", "--warmup", "0")
     assert REQUIRED_KEYS <= set(out)
     assert out["metric"].startswith("shap-configs/sec")
     assert len(out["config"]["shap_configs"]) == 2

@@ -262,7 +262,7 @@ class TestFiguresStage:
             COMPARISON_CONFIGS, write_figures,
         )
 
-        tests = make_synthetic_tests(n_tests=300, seed=7)
+        tests = make_synthetic_tests(n_tests=200, seed=7)
         tests_file = tmp_path / "tests.json"
         with open(tests_file, "w") as fd:
             json.dump(tests, fd)
