@@ -110,9 +110,11 @@ def shard_cells(world, rank, n_cells=None):
 
 def _pack(result, cell_order, projects):
     """result {config_keys: [t_train, t_test, scores, scores_total]} ->
-    float64 tensor [n_cells, 2 + (n_proj+1)*3]."""
+    float64 tensor [n_cells, 2 + (n_proj+1)*3 + 1]; the trailing column
+    marks cells this rank actually evaluated (so a partial-grid run does
+    not gather phantom zero-count cells)."""
     n_proj = len(projects)
-    buf = np.zeros((len(cell_order), 2 + (n_proj + 1) * _N_CONFUSION))
+    buf = np.zeros((len(cell_order), 2 + (n_proj + 1) * _N_CONFUSION + 1))
     for ci, keys in enumerate(cell_order):
         if keys not in result:
             continue
@@ -121,7 +123,8 @@ def _pack(result, cell_order, projects):
         buf[ci, 1] = t_test
         for pi, proj in enumerate(projects):
             buf[ci, 2 + pi * 3: 2 + pi * 3 + 3] = scores[proj][:3]
-        buf[ci, 2 + n_proj * 3:] = scores_total[:3]
+        buf[ci, 2 + n_proj * 3: 2 + (n_proj + 1) * 3] = scores_total[:3]
+        buf[ci, -1] = 1.0
     return buf
 
 
@@ -129,10 +132,13 @@ def _unpack(buf, cell_order, projects):
     n_proj = len(projects)
     out = {}
     for ci, keys in enumerate(cell_order):
+        if buf[ci, -1] == 0.0:
+            continue   # no rank evaluated this cell
         scores = {}
         for pi, proj in enumerate(projects):
             scores[proj] = [int(v) for v in buf[ci, 2 + pi * 3: 2 + pi * 3 + 3]]
-        scores_total = [int(v) for v in buf[ci, 2 + n_proj * 3:]]
+        scores_total = [int(v)
+                        for v in buf[ci, 2 + n_proj * 3: 2 + (n_proj + 1) * 3]]
         finalize_scores(scores, scores_total)
         out[keys] = [float(buf[ci, 0]), float(buf[ci, 1]), scores, scores_total]
     return out

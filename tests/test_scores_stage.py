@@ -98,7 +98,8 @@ WORKER = textwrap.dedent("""
     dist.init_process_group("gloo")
     rank, world = comm.rank_world()
     tests = make_synthetic_tests(n_tests=300, seed=2)
-    cells = comm.shard_cells(world, rank, n_cells=6)
+    n_cells = int(os.environ.get("TEST_N_CELLS", "6"))
+    cells = comm.shard_cells(world, rank, n_cells=n_cells)
     result = run_scores(tests=tests, backend="ref", cells=cells)
     full = comm.gather_scores(result)
     if rank == 0:
@@ -109,6 +110,37 @@ WORKER = textwrap.dedent("""
 
 
 class TestDistributedGloo:
+    def _launch(self, tmp_path, n_cells, port):
+        out = str(tmp_path / "dist_scores.pkl")
+        script = tmp_path / "worker.py"
+        script.write_text(WORKER.format(repo="/root/repo", out=out))
+        env = dict(os.environ, MASTER_ADDR="127.0.0.1",
+                   MASTER_PORT=str(port), TEST_N_CELLS=str(n_cells))
+        procs = []
+        for rank in range(2):
+            env_r = dict(env, RANK=str(rank), WORLD_SIZE="2",
+                         LOCAL_RANK=str(rank))
+            procs.append(subprocess.Popen(
+                [sys.executable, str(script)], env=env_r,
+                stdout=subprocess.PIPE, stderr=subprocess.STDOUT))
+        for p in procs:
+            outp, _ = p.communicate(timeout=300)
+            assert p.returncode == 0, outp.decode()
+        with open(out, "rb") as fd:
+            return pickle.load(fd)
+
+    def test_empty_rank_gathers_clean(self, tmp_path):
+        """n_cells=3 puts the whole (single) balance group on rank 0;
+        rank 1 owns nothing — the all-reduce must still produce the
+        full result (guards the empty-shard path at world=8)."""
+        dist_result = self._launch(tmp_path, n_cells=3, port=29612)
+        tests = make_synthetic_tests(n_tests=300, seed=2)
+        single = run_scores(tests=tests, backend="ref", cells=[0, 1, 2])
+        assert set(dist_result) == set(single)
+        for k in single:
+            assert dist_result[k][2] == single[k][2]
+            assert dist_result[k][3] == single[k][3]
+
     def test_two_rank_shard_equals_single(self, tmp_path):
         out = str(tmp_path / "dist_scores.pkl")
         script = tmp_path / "worker.py"

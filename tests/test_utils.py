@@ -68,9 +68,10 @@ class TestCommPack:
         # PRF recomputed: P = 9/(9+5), R = 9/(9+7)
         assert abs(got[3][3] - 9 / 14) < 1e-12
         assert abs(got[3][4] - 9 / 16) < 1e-12
-        # unowned cells come back as zero counts with None PRF
-        assert out[cell_order[0]][3][:3] == [0, 0, 0]
-        assert out[cell_order[0]][3][3] is None
+        # unevaluated cells are absent (presence column), not phantom
+        # zero-count entries
+        assert cell_order[0] not in out
+        assert set(out) == {cell_order[1]}
 
 
 class TestTrace:
