@@ -176,12 +176,12 @@ class TestDistributedGloo:
             all_cells = sorted(
                 c for r in range(world) for c in shard_cells(world, r))
             assert all_cells == list(range(216))
-        # load is balanced within ~30% across 8 ranks
+        # LPT balance: within ~12% across 8 ranks on the cost model
         from flake16_framework_amd.configgrid import cell_cost_estimate
         keys = ALL_KEYS
         loads = [sum(cell_cost_estimate(keys[c]) for c in shard_cells(8, r))
                  for r in range(8)]
-        assert max(loads) / min(loads) < 1.3
+        assert max(loads) / min(loads) < 1.12
 
     def test_ref_process_pool_matches_serial(self):
         """run_scores(processes=2) — the reference's Pool execution model
