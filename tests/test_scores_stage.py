@@ -109,6 +109,20 @@ WORKER = textwrap.dedent("""
 """)
 
 
+SHAP_WORKER = textwrap.dedent("""
+    import os, pickle, sys
+    import torch.distributed as dist
+    sys.path.insert(0, {repo!r})
+    from flake16_framework_amd.dataset.synthetic import make_synthetic_tests
+    from flake16_framework_amd.engine.shap_stage import write_shap
+
+    dist.init_process_group("gloo")
+    tests = make_synthetic_tests(n_tests=150, seed=2)
+    write_shap(tests=tests, shap_file={out!r}, backend="ref")
+    dist.destroy_process_group()
+""")
+
+
 class TestDistributedGloo:
     def _launch(self, tmp_path, n_cells, port):
         out = str(tmp_path / "dist_scores.pkl")
@@ -140,6 +154,35 @@ class TestDistributedGloo:
         for k in single:
             assert dist_result[k][2] == single[k][2]
             assert dist_result[k][3] == single[k][3]
+
+    def test_shap_rank_split_equals_single(self, tmp_path):
+        """write_shap splits the 2 configs across ranks and all-gathers;
+        the merged shap.pkl must equal the single-process result."""
+        out = str(tmp_path / "shap.pkl")
+        script = tmp_path / "shap_worker.py"
+        script.write_text(SHAP_WORKER.format(repo="/root/repo", out=out))
+        env = dict(os.environ, MASTER_ADDR="127.0.0.1",
+                   MASTER_PORT="29613")
+        procs = []
+        for rank in range(2):
+            env_r = dict(env, RANK=str(rank), WORLD_SIZE="2",
+                         LOCAL_RANK=str(rank))
+            procs.append(subprocess.Popen(
+                [sys.executable, str(script)], env=env_r,
+                stdout=subprocess.PIPE, stderr=subprocess.STDOUT))
+        for p in procs:
+            outp, _ = p.communicate(timeout=300)
+            assert p.returncode == 0, outp.decode()
+
+        from flake16_framework_amd.engine.shap_stage import write_shap
+        tests = make_synthetic_tests(n_tests=150, seed=2)
+        single = write_shap(tests=tests,
+                            shap_file=str(tmp_path / "single.pkl"),
+                            backend="ref")
+        with open(out, "rb") as fd:
+            dist_nod, dist_od = pickle.load(fd)
+        np.testing.assert_array_equal(dist_nod, single[0])
+        np.testing.assert_array_equal(dist_od, single[1])
 
     def test_two_rank_shard_equals_single(self, tmp_path):
         out = str(tmp_path / "dist_scores.pkl")
